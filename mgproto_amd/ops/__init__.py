@@ -1,0 +1,147 @@
+"""Prototype-math ops with CPU (PyTorch) and MI355X (HIP/CDNA4) paths.
+
+Dispatch policy:
+* CPU tensors -> the pure-PyTorch reference implementations (the oracle).
+* CUDA (ROCm) tensors -> the in-tree ``mgproto_hip`` extension (gfx950
+  MFMA kernels). If the extension is missing on a GPU box this raises —
+  a silent eager fallback would invalidate benchmarks. Set
+  ``MGPROTO_EAGER_FALLBACK=1`` to explicitly allow the PyTorch path on GPU
+  (used for A/B measurements only).
+"""
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference
+from .reference import (gmm_expand_params, gmm_logprob_direct, mask_wrong_class,  # noqa: F401
+                        mixture_head, gather_patch_features, enqueue_candidates,
+                        em_e_step, em_m_step_grads)
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_extension():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from . import hip_loader
+        _EXT = hip_loader.load()
+    except Exception as e:  # noqa: BLE001
+        _EXT_ERR = f'{type(e).__name__}: {e}'
+        _EXT = None
+    return _EXT
+
+
+def native_available() -> bool:
+    return _load_extension() is not None
+
+
+def _native_or_die():
+    ext = _load_extension()
+    if ext is None:
+        if os.environ.get('MGPROTO_EAGER_FALLBACK') == '1':
+            return None
+        raise RuntimeError(
+            'mgproto_hip extension is required for GPU execution but could not '
+            f'be loaded ({_EXT_ERR}). Build it with '
+            '`python -m mgproto_amd.ops.build` (or __graft_entry__.build()), '
+            'or set MGPROTO_EAGER_FALLBACK=1 to explicitly allow the slow '
+            'PyTorch path.')
+    return ext
+
+
+class _GMMScore(torch.autograd.Function):
+    """Fused GMM score GEMM: out[n,p] = f(bias[p] + [x, x^2] @ W), f = exp or id.
+
+    Forward/backward run on the gfx950 MFMA kernels for CUDA tensors; grads
+    flow to ``feat`` only (means/covs are detached by the caller, matching
+    reference model.py:264-265).
+    """
+
+    @staticmethod
+    def forward(ctx, feat, W, bias, apply_exp):
+        ext = _native_or_die() if feat.is_cuda else None
+        if ext is not None:
+            out = ext.gmm_fwd(feat, W, bias, apply_exp)
+        else:
+            out = torch.addmm(bias.unsqueeze(0),
+                              torch.cat([feat, feat * feat], dim=1), W)
+            if apply_exp:
+                out = torch.exp(out)
+        ctx.save_for_backward(feat, W, out if apply_exp else torch.empty(0))
+        ctx.apply_exp = apply_exp
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        feat, W, probs = ctx.saved_tensors
+        g = grad_out * probs if ctx.apply_exp else grad_out
+        ext = _native_or_die() if feat.is_cuda else None
+        if ext is not None:
+            grad_feat = ext.gmm_bwd(g.contiguous(), feat, W)
+        else:
+            gw = g @ W.t()                       # [N, 2d]
+            d = feat.shape[1]
+            grad_feat = gw[:, :d] + 2.0 * feat * gw[:, d:]
+        return grad_feat, None, None, None
+
+
+def gmm_scores(feat: torch.Tensor, means: torch.Tensor, covs: torch.Tensor,
+               apply_exp: bool = True, eps: float = 0.0) -> torch.Tensor:
+    """[N, d] patch features -> [N, P] Gaussian (log-)likelihoods.
+
+    The hot K1 op (reference model.py:213-215). ``apply_exp=True`` returns
+    probabilities (forward path); ``False`` returns log-probabilities
+    (`compute_log_prob` API parity).
+    """
+    means2 = means.reshape(-1, means.shape[-1]).detach()
+    covs2 = covs.reshape(-1, covs.shape[-1]).detach()
+    W, bias = gmm_expand_params(means2.float(), covs2.float(), eps)
+    return _GMMScore.apply(feat, W, bias, apply_exp)
+
+
+class _TopkHW(torch.autograd.Function):
+    """Per-(image, prototype) top-T over the spatial axis, with indices."""
+
+    @staticmethod
+    def forward(ctx, probs, T):
+        ext = _native_or_die() if probs.is_cuda else None
+        if ext is not None:
+            vals, idx = ext.topk_hw(probs, T)
+        else:
+            v, i = torch.topk(probs, T, dim=1)
+            vals, idx = v.permute(0, 2, 1).contiguous(), i.permute(0, 2, 1).contiguous()
+        ctx.save_for_backward(idx)
+        ctx.hw = probs.shape[1]
+        return vals, idx.to(torch.long)
+
+    @staticmethod
+    def backward(ctx, grad_vals, _grad_idx):
+        (idx,) = ctx.saved_tensors
+        B, P, T = idx.shape
+        grad_probs = grad_vals.new_zeros(B, ctx.hw, P)
+        # scatter-add grads back to their source patches
+        grad_probs.scatter_add_(1, idx.to(torch.long).permute(0, 2, 1),
+                                grad_vals.permute(0, 2, 1))
+        return grad_probs, None
+
+
+def topk_hw(probs: torch.Tensor, T: int) -> Tuple[torch.Tensor, torch.Tensor]:
+    """[B, HW, P] -> (values [B, P, T] desc-sorted, indices [B, P, T])."""
+    return _TopkHW.apply(probs, T)
+
+
+def argmax_hw(probs: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """[B, HW, P] -> (max values [B, P], hw indices [B, P]). No autograd
+    (push path runs under no_grad). Lowest-index-wins on ties."""
+    if probs.is_cuda:
+        ext = _native_or_die()
+        if ext is not None:
+            vals, idx = ext.argmax_hw(probs)
+            return vals, idx.to(torch.long)
+    vals, idx = probs.max(dim=1)
+    return vals, idx

@@ -1,0 +1,60 @@
+"""Small shared helpers (reference ``utils/helpers.py``), cv2-free.
+
+``find_high_activation_crop`` uses scipy.ndimage connected components instead
+of OpenCV (not installed here); semantics match reference helpers.py:38-74.
+"""
+
+import os
+import time
+
+import numpy as np
+import torch
+
+try:
+    from scipy import ndimage
+except ImportError:  # pragma: no cover
+    ndimage = None
+
+
+def list_of_distances(X: torch.Tensor, Y: torch.Tensor) -> torch.Tensor:
+    """Pairwise squared L2 distances [n, m] (reference helpers.py:13)."""
+    return torch.sum((torch.unsqueeze(X, dim=2) - torch.unsqueeze(Y.t(), dim=0)) ** 2,
+                     dim=1)
+
+
+def make_one_hot(target, target_one_hot):
+    target = target.view(-1, 1)
+    target_one_hot.zero_()
+    target_one_hot.scatter_(dim=1, index=target, value=1.)
+
+
+def makedir(path):
+    if not os.path.exists(path):
+        os.makedirs(path, exist_ok=True)
+
+
+def datestr():
+    now = time.gmtime()
+    return '{}{:02}{:02}_{:02}{:02}'.format(now.tm_year, now.tm_mon, now.tm_mday,
+                                            now.tm_hour, now.tm_min)
+
+
+def find_high_activation_crop(activation_map: np.ndarray, percentile: float = 95):
+    """Bounding box of the connected high-activation component containing the
+    activation peak (reference helpers.py:38-74, scipy instead of cv2)."""
+    threshold = np.percentile(activation_map, percentile)
+    mask = (activation_map >= threshold).astype(np.uint8)
+
+    hi = np.unravel_index(np.argmax(activation_map), activation_map.shape)
+    if ndimage is not None:
+        labeled, n_labels = ndimage.label(mask, structure=np.ones((3, 3)))
+        peak_label = labeled[hi[0], hi[1]]
+        if peak_label > 0:
+            mask = (labeled == peak_label).astype(np.uint8)
+
+    ys, xs = np.where(mask > 0)
+    if len(ys) == 0:
+        return (hi[0], hi[0] + 1, hi[1], hi[1] + 1)
+    lower_y, upper_y = int(ys.min()), int(ys.max())
+    lower_x, upper_x = int(xs.min()), int(xs.max())
+    return (lower_y, upper_y + 1, lower_x, upper_x + 1)

@@ -1,0 +1,180 @@
+"""Class-wise FIFO memory bank of patch features.
+
+Re-design of the reference MemoryBank (``/root/reference/utils/memory.py``):
+
+* storage is ONE device-resident ``[C, cap, d]`` buffer + per-class length
+  and ring head, instead of 200 separate ``cls{i}`` buffers — one scatter
+  writes a whole batch (the reference loops classes in Python,
+  memory.py:48-71, and shifts the whole buffer on overflow, memory.py:64);
+* ``push`` is fully batched and deterministic (stable sort by class, FIFO
+  ring positions) with no host synchronization;
+* the state-dict surface still presents the reference layout
+  (``cls0..cls{C-1}`` of shape [cap, d] in oldest-first order plus
+  ``mem_len``), so checkpoints interoperate both ways.
+
+Semantics relative to the reference: the stored multiset per class is
+identical (FIFO, capacity cap); rows may sit at different physical offsets
+(ring vs shift) which is unobservable through ``pull_*``/state_dict (both
+return logical order). One divergence: when a single push exceeds cap the
+reference keeps a *random* subsample (memory.py:52) — we keep the newest
+cap items (deterministic; unreachable with default shapes since
+B*K <= cap).
+"""
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+
+class MemoryBank(nn.Module):
+    def __init__(self, num_classes: int, dim_feature: int, capacity: int = 1024,
+                 mode: str = 'all', fix_length_mult: int = 4):
+        super().__init__()
+        assert capacity % num_classes == 0, (capacity, num_classes)
+        self.num_classes = num_classes
+        self.dim_feature = dim_feature
+        self.capacity = capacity
+        self.cap_cls = capacity // num_classes
+        self.mode = mode
+        self.fix_length_mult = fix_length_mult
+
+        self.register_buffer('mem', torch.zeros(num_classes, self.cap_cls,
+                                                dim_feature))
+        self.register_buffer('mem_len', torch.zeros(num_classes, dtype=torch.int64))
+        self.register_buffer('head', torch.zeros(num_classes, dtype=torch.int64))
+
+    # ------------------------------------------------------------------ push
+    @torch.no_grad()
+    def push(self, feature: torch.Tensor, label: torch.Tensor) -> None:
+        """Append (feature, label) pairs FIFO per class. Batched, in-order.
+
+        ``feature``: [M, d]; ``label``: [M] int64. Items of one class are
+        inserted in their order of appearance.
+        """
+        assert feature.dim() == 2 and label.dim() == 1
+        assert feature.size(0) == label.size(0)
+        M = feature.size(0)
+        if M == 0:
+            return
+        feature = feature.detach()
+        label = label.detach()
+        cap = self.cap_cls
+        dev = feature.device
+
+        order = torch.argsort(label, stable=True)
+        lab = label[order]
+        fea = feature[order]
+
+        # position within each class segment
+        change = torch.ones(M, dtype=torch.bool, device=dev)
+        change[1:] = lab[1:] != lab[:-1]
+        seg_id = torch.cumsum(change.long(), 0) - 1                 # [M]
+        idx = torch.arange(M, device=dev)
+        seg_first = idx[change]                                     # [#segments]
+        within = idx - seg_first[seg_id]                            # [M]
+
+        counts = torch.zeros(self.num_classes, dtype=torch.int64, device=dev)
+        counts.scatter_add_(0, lab, torch.ones_like(lab))
+
+        # if a class pushes more than cap at once, keep only the newest cap
+        # (positions keep their original within-offsets so the advanced head
+        # still points at the oldest surviving row)
+        keep = within >= (counts[lab] - cap)
+        lab, fea, within = lab[keep], fea[keep], within[keep]
+
+        pos = (self.head[lab] + within) % cap
+        self.mem[lab, pos] = fea.to(self.mem.dtype)
+        self.head = (self.head + counts) % cap
+        self.mem_len = torch.minimum(self.mem_len + counts,
+                                     torch.full_like(self.mem_len, cap))
+
+    # ------------------------------------------------------------------ pull
+    def _logical(self, c: int) -> torch.Tensor:
+        """Rows of class c, oldest first."""
+        L = int(self.mem_len[c])
+        h = int(self.head[c])
+        if L < self.cap_cls:
+            return self.mem[c, :L]
+        return torch.cat([self.mem[c, h:], self.mem[c, :h]], dim=0)
+
+    @torch.no_grad()
+    def pull_all(self) -> Tuple[Optional[torch.Tensor], Optional[torch.Tensor]]:
+        """All stored features + labels (reference memory.py:136-151)."""
+        lens = self.mem_len
+        if int(lens.sum()) == 0:
+            return None, None
+        out_data, out_label = [], []
+        for c in range(self.num_classes):
+            L = int(lens[c])
+            if L == 0:
+                continue
+            out_data.append(self._logical(c))
+            out_label.append(torch.full((L,), c, dtype=torch.int64,
+                                        device=self.mem.device))
+        return torch.cat(out_data, 0), torch.cat(out_label, 0)
+
+    @torch.no_grad()
+    def pull_dense(self, classes: torch.Tensor) -> torch.Tensor:
+        """[G, cap, d] stack for the given (full) classes — the batched-EM
+        feed. Requires every requested class to be full; physical order is
+        used (EM is permutation-invariant over samples)."""
+        return self.mem.index_select(0, classes)
+
+    def full_mask(self) -> torch.Tensor:
+        return self.mem_len == self.cap_cls
+
+    @torch.no_grad()
+    def pull_fix_class(self, label: torch.Tensor):
+        """Reference memory.py:76-93: features of the flagged classes."""
+        assert list(label.size()) == [self.num_classes]
+        indices = torch.nonzero(label.detach(), as_tuple=False).flatten()
+        out_data, out_label = [], []
+        for i in indices.tolist():
+            L = int(self.mem_len[i])
+            if L == 0:
+                continue
+            out_data.append(self._logical(i))
+            out_label.append(torch.full((L,), i, dtype=torch.int64,
+                                        device=self.mem.device))
+        if not out_data:
+            return None, None
+        return torch.cat(out_data, 0), torch.cat(out_label, 0)
+
+    def pull(self, *args, **kwargs):
+        if self.mode == 'all':
+            return self.pull_all(*args, **kwargs)
+        return getattr(self, 'pull_' + self.mode)(*args, **kwargs)
+
+    # ------------------------------------------------- reference state_dict
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        # Export the reference layout: cls{i} [cap, d] oldest-first + mem_len.
+        for c in range(self.num_classes):
+            buf = torch.zeros_like(self.mem[c])
+            L = int(self.mem_len[c])
+            if L > 0:
+                buf[:L] = self._logical(c)[:L] if L < self.cap_cls else self._logical(c)
+            destination[prefix + 'cls%d' % c] = buf
+        destination[prefix + 'mem_len'] = (self.mem_len if keep_vars
+                                           else self.mem_len.detach().clone())
+
+    def _load_from_state_dict(self, state_dict, prefix, local_metadata, strict,
+                              missing_keys, unexpected_keys, error_msgs):
+        found = 0
+        for c in range(self.num_classes):
+            key = prefix + 'cls%d' % c
+            if key in state_dict:
+                self.mem[c].copy_(state_dict[key])
+                found += 1
+                state_dict.pop(key)
+        key = prefix + 'mem_len'
+        if key in state_dict:
+            self.mem_len.copy_(state_dict[key])
+            state_dict.pop(key)
+        elif strict:
+            missing_keys.append(key)
+        # logical order was exported oldest-first, so heads reset
+        self.head.copy_(self.mem_len % self.cap_cls)
+        if strict and found not in (0, self.num_classes):
+            error_msgs.append(f'MemoryBank: only {found}/{self.num_classes} '
+                              'cls buffers present in checkpoint')
