@@ -1,0 +1,300 @@
+"""Training / evaluation engine.
+
+Same control flow as the reference (``/root/reference/train_and_test.py``):
+per-batch forward, CE + mine-CE + aux loss, backward/step, in-loop EM
+trigger, accuracy bookkeeping; test loop; OoD variant — re-designed for
+MI355X execution:
+
+* bf16 autocast over the backbone (prototype math stays fp32), NHWC layout;
+* no per-batch host syncs — loss/accuracy stats accumulate in device
+  tensors and only materialize at print points / epoch end (the reference
+  calls ``.item()`` per batch, train_and_test.py:50-52);
+* the EM update runs on a side HIP stream overlapped with the next step's
+  backbone (it is state, not gradient — SURVEY.md §5);
+* distributed: stats are all-reduced (C4); the model's enqueue/EM are
+  already DP-correct via mgproto_amd.parallel.
+"""
+
+import contextlib
+import time
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from ..utils.helpers import list_of_distances
+
+
+def _unwrap(model):
+    return model.module if hasattr(model, 'module') else model
+
+
+def _amp_ctx(device, amp_dtype):
+    if device.type == 'cuda' and amp_dtype in ('bf16', 'fp16'):
+        dt = torch.bfloat16 if amp_dtype == 'bf16' else torch.float16
+        return torch.autocast(device_type='cuda', dtype=dt)
+    return contextlib.nullcontext()
+
+
+class EMRunner:
+    """Runs update_GMM, optionally on a side stream overlapped with the next
+    step's backbone work. The consumer (gmm_scores in the next forward) waits
+    on the recorded event via ``sync()``."""
+
+    def __init__(self, model, use_stream: bool):
+        self.model = model
+        self.use_stream = use_stream and torch.cuda.is_available()
+        self.stream = torch.cuda.Stream() if self.use_stream else None
+        self.event = torch.cuda.Event() if self.use_stream else None
+        self._pending = False
+
+    def run(self):
+        m = _unwrap(self.model)
+        if not self.use_stream:
+            m.update_GMM()
+            return
+        self.stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self.stream):
+            m.update_GMM()
+        self.event.record(self.stream)
+        self._pending = True
+
+    def sync(self):
+        if self._pending:
+            torch.cuda.current_stream().wait_event(self.event)
+            self._pending = False
+
+
+def _training(model, dataloader, optimizer=None, aux_criterion=None,
+              use_mine=False, update_GMM=False, class_specific=True,
+              coefs=None, log=print, device=None, amp_dtype='bf16',
+              em_runner: Optional[EMRunner] = None, metrics=None,
+              comm=None, print_every=20):
+    device = device or next(_unwrap(model).parameters()).device
+    m = _unwrap(model)
+    start = time.time()
+
+    n_examples = torch.zeros((), device=device)
+    n_correct = torch.zeros((), device=device)
+    total_ce = torch.zeros((), device=device)
+    total_mine = torch.zeros((), device=device)
+    total_aux = torch.zeros((), device=device)
+    n_batches = 0
+
+    for i, batch in enumerate(dataloader):
+        image, label = batch[0], batch[1]
+        image = image.to(device, non_blocking=True)
+        target = label.to(device, non_blocking=True)
+
+        if em_runner is not None:
+            em_runner.sync()  # next forward reads EM-updated means/priors
+
+        with _amp_ctx(device, amp_dtype):
+            output, x_auxiliary = model(image, target)
+
+        output = output.float()
+        if use_mine and output.shape[2] > 1:
+            mine_loss = sum(F.cross_entropy(output[:, :, k], target)
+                            for k in range(1, output.shape[2])) / (output.shape[2] - 1)
+        else:
+            mine_loss = torch.zeros((), device=device)
+        cross_entropy = F.cross_entropy(output[:, :, 0], target)
+        aux_loss = (aux_criterion(x_auxiliary.float(), target)
+                    if aux_criterion is not None else torch.zeros((), device=device))
+
+        predicted = torch.argmax(output[:, :, 0].detach(), dim=1)
+        n_examples += target.numel()
+        n_correct += (predicted == target).sum()
+        n_batches += 1
+        total_ce += cross_entropy.detach()
+        total_mine += mine_loss.detach()
+        total_aux += aux_loss.detach()
+
+        loss = (coefs['crs_ent'] * cross_entropy + coefs['mine'] * mine_loss
+                + coefs['aux'] * aux_loss)
+        optimizer.zero_grad(set_to_none=True)
+        loss.backward()
+        optimizer.step()
+
+        # EM update (reference train_and_test.py:61-63; update_interval=1)
+        if update_GMM and int(m.queue.mem_len.sum()) > 0:
+            if int(m.iteration_counter) % m.update_interval == 0:
+                (em_runner.run() if em_runner is not None else m.update_GMM())
+
+        if print_every and i % print_every == 0:
+            full_ratio = (m.queue.mem_len == m.capacity_pc).float().mean()
+            acc = float(n_correct) / (float(n_examples) + 1e-6) * 100
+            print(f'{i} {len(dataloader)} \tLoss: {float(loss):.4f} '
+                  f'\tL_ce: {float(cross_entropy):.4f} '
+                  f'\tL_mine: {float(mine_loss):.4f} '
+                  f'\tL_aux: {float(aux_loss):.4f} '
+                  f'\tMem_ratio {float(full_ratio):.2f} \tAcc: {acc:.4f}')
+            if metrics is not None:
+                metrics.log({'train/loss': float(loss),
+                             'train/ce': float(cross_entropy),
+                             'train/mine': float(mine_loss),
+                             'train/aux': float(aux_loss),
+                             'train/acc': acc,
+                             'train/mem_ratio': float(full_ratio)})
+
+    if em_runner is not None:
+        em_runner.sync()
+
+    stats = torch.stack([n_correct, n_examples, total_ce, total_mine, total_aux])
+    if comm is not None:
+        stats = comm.all_reduce_sum(stats)
+    n_correct, n_examples, total_ce, total_mine, total_aux = stats.tolist()
+    nb = n_batches * (comm.world_size if comm is not None else 1)
+
+    log('\ttime: \t{0}'.format(time.time() - start))
+    log('\tcross ent: \t{0}'.format(total_ce / max(nb, 1)))
+    log('\tmine: \t{0}'.format(total_mine / max(nb, 1)))
+    log('\taux: \t{0}'.format(total_aux / max(nb, 1)))
+
+    results = {'cross_entropy': total_ce / max(nb, 1),
+               'mine_loss': total_mine / max(nb, 1),
+               'aux_loss': total_aux / max(nb, 1),
+               'acc': n_correct / max(n_examples, 1)}
+    return results['acc'], results
+
+
+@torch.no_grad()
+def _testing(model, dataloader, class_specific=True, log=print, device=None,
+             amp_dtype='bf16', comm=None, metrics=None):
+    device = device or next(_unwrap(model).parameters()).device
+    m = _unwrap(model)
+    start = time.time()
+    n_examples = torch.zeros((), device=device)
+    n_correct = torch.zeros((), device=device)
+    total_ce = torch.zeros((), device=device)
+    n_batches = 0
+
+    loader = dataloader[0] if isinstance(dataloader, (tuple, list)) else dataloader
+    for batch in loader:
+        image, label = batch[0], batch[1]
+        image = image.to(device, non_blocking=True)
+        target = label.to(device, non_blocking=True)
+        with _amp_ctx(device, amp_dtype):
+            output, _ = model(image, None)
+        output = output.float()
+        total_ce += F.cross_entropy(output[:, :, 0], target)
+        predicted = torch.argmax(output[:, :, 0], dim=1)
+        n_examples += target.numel()
+        n_correct += (predicted == target).sum()
+        n_batches += 1
+
+    stats = torch.stack([n_correct, n_examples, total_ce])
+    if comm is not None:
+        stats = comm.all_reduce_sum(stats)
+    n_correct, n_examples, total_ce = stats.tolist()
+    nb = n_batches * (comm.world_size if comm is not None else 1)
+
+    log('\ttime: \t{0}'.format(time.time() - start))
+    log('\tcross ent: \t{0}'.format(total_ce / max(nb, 1)))
+    log('\ttest acc: \t\t{0}%'.format(n_correct / max(n_examples, 1) * 100))
+
+    p = m.prototype_means.view(m.num_prototypes, -1).cpu()
+    p_avg_pair_dist = torch.mean(list_of_distances(p, p))
+    log('\tp dist pair: \t{0}'.format(p_avg_pair_dist.item()))
+    if metrics is not None:
+        metrics.log({'test/acc': n_correct / max(n_examples, 1) * 100})
+
+    results = {'cross_entropy': total_ce / max(nb, 1),
+               'p_avg_pair_dist': p_avg_pair_dist,
+               'acc': n_correct / max(n_examples, 1)}
+    return results['acc'], results
+
+
+@torch.no_grad()
+def _testing_with_OoD(model, dataloaders, class_specific=True, log=print,
+                      device=None, amp_dtype='bf16', comm=None, metrics=None,
+                      percentile=5):
+    """ID accuracy + OoD FPR95 via the mixture density p(x)
+    (reference train_and_test.py:163-242)."""
+    device = device or next(_unwrap(model).parameters()).device
+    id_loader, ood_loaders = dataloaders[0], dataloaders[1:]
+
+    n_examples = torch.zeros((), device=device)
+    n_correct = torch.zeros((), device=device)
+    id_probs = []
+    for batch in id_loader:
+        image, label = batch[0], batch[1]
+        image = image.to(device, non_blocking=True)
+        target = label.to(device, non_blocking=True)
+        with _amp_ctx(device, amp_dtype):
+            output, _ = model(image, None)
+        output_prob = output[:, :, 0].float().exp()          # p(x, c)
+        id_probs.append(output_prob.sum(dim=1))              # p(x)
+        predicted = torch.argmax(output[:, :, 0], dim=1)
+        n_examples += target.numel()
+        n_correct += (predicted == target).sum()
+
+    id_probs = torch.cat(id_probs)
+    if comm is not None:
+        id_probs = comm.all_gather_varlen(id_probs)
+        stats = comm.all_reduce_sum(torch.stack([n_correct, n_examples]))
+        n_correct, n_examples = stats.tolist()
+    else:
+        n_correct, n_examples = float(n_correct), float(n_examples)
+    acc = n_correct / max(n_examples, 1) * 100
+    log('\tTest Acc: \t{0}'.format(acc))
+    ood_thresh = torch.quantile(id_probs.float().cpu(), percentile / 100.0)
+
+    results = {'acc': acc}
+    for li, loader in enumerate(ood_loaders, start=1):
+        preds = []
+        for batch in loader:
+            image = batch[0].to(device, non_blocking=True)
+            with _amp_ctx(device, amp_dtype):
+                output, _ = model(image, None)
+            output_prob = output[:, :, 0].float().exp()
+            # reference :213 thresholds the class-MEAN density
+            preds.append(output_prob.mean(dim=1) > ood_thresh.to(device))
+        preds = torch.cat(preds) if preds else torch.zeros(0, device=device)
+        if comm is not None:
+            preds = comm.all_gather_varlen(preds.float())
+        fpr95 = float(preds.float().sum()) / max(preds.numel(), 1)
+        log('\tFPR95_{0}: \t{1}'.format(li, fpr95))
+        results[f'FPR95_{li}'] = fpr95
+        if metrics is not None:
+            metrics.log({f'ood/FPR95_{li}': fpr95})
+    return n_correct / max(n_examples, 1), results
+
+
+def train(model, dataloader, optimizer, aux_criterion=None, use_mine=False,
+          update_GMM=False, class_specific=False, coefs=None, log=print,
+          **kw):
+    assert optimizer is not None
+    log('\ttrain')
+    model.train()
+    return _training(model=model, dataloader=dataloader, optimizer=optimizer,
+                     aux_criterion=aux_criterion, use_mine=use_mine,
+                     update_GMM=update_GMM, class_specific=class_specific,
+                     coefs=coefs, log=log, **kw)
+
+
+def test(model, dataloader, class_specific=False, log=print, ood=False, **kw):
+    log('\ttest')
+    model.eval()
+    if ood:
+        return _testing_with_OoD(model, dataloader, class_specific=class_specific,
+                                 log=log, **kw)
+    return _testing(model, dataloader, class_specific=class_specific, log=log, **kw)
+
+
+def warm_only(model, log=print):
+    m = _unwrap(model)
+    for p in m.features.parameters():
+        p.requires_grad = False
+    for p in m.add_on_layers.parameters():
+        p.requires_grad = True
+    log('\twarm')
+
+
+def joint(model, log=print):
+    m = _unwrap(model)
+    for p in m.features.parameters():
+        p.requires_grad = True
+    for p in m.add_on_layers.parameters():
+        p.requires_grad = True
+    log('\tjoint')

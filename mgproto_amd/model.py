@@ -240,6 +240,12 @@ class MGProto(nn.Module):
     # -------------------------------------------------------------- forward
     def forward(self, x, gt):
         base_feature, x_auxiliary = self.conv_features(x)
+        # prototype math runs fp32 even when the backbone is under bf16
+        # autocast (the engine autocasts only the conv stack)
+        with torch.autocast(device_type=base_feature.device.type, enabled=False):
+            return self._prototype_forward(base_feature, x_auxiliary, gt)
+
+    def _prototype_forward(self, base_feature, x_auxiliary, gt):
         base_feature = l2_normalize(base_feature.float(), dim=1)
         B, d, H, W = base_feature.shape
         HW = H * W
@@ -291,11 +297,12 @@ class MGProto(nn.Module):
         Reference model.py:429-438: distances = -exp(log_prob).
         """
         base_feature, _ = self.conv_features(x)
-        base_feature = l2_normalize(base_feature.float(), dim=1)
-        B, d, H, W = base_feature.shape
-        feat = base_feature.permute(0, 2, 3, 1).reshape(B * H * W, d)
-        probs = ops.gmm_scores(feat, self.prototype_means, self.prototype_covs,
-                               apply_exp=True)
+        with torch.autocast(device_type=base_feature.device.type, enabled=False):
+            base_feature = l2_normalize(base_feature.float(), dim=1)
+            B, d, H, W = base_feature.shape
+            feat = base_feature.permute(0, 2, 3, 1).reshape(B * H * W, d)
+            probs = ops.gmm_scores(feat, self.prototype_means,
+                                   self.prototype_covs, apply_exp=True)
         distances = -probs.view(B, H * W, self.num_prototypes) \
                           .permute(0, 2, 1).reshape(B, self.num_prototypes, H, W)
         return base_feature, distances

@@ -1,0 +1,438 @@
+// MGProto prototype-math kernels for MI355X (gfx950, CDNA4).
+//
+// Implements the hot ops identified in SURVEY.md §2.2 as native HIP:
+//   K1  gmm_fwd : out[n,p] = f( bias[p] + sum_j x[n,j]*W[j,p] + x[n,j]^2*W[d+j,p] )
+//                 with f = exp (fused epilogue) or identity.
+//                 (reference model.py:256-275 + .exp() at :215, recast as one
+//                  [N,2d]x[2d,P] GEMM on fp32 MFMA — exact f32 numerics,
+//                  v_mfma_f32_16x16x4_f32.)
+//   K1b gmm_bwd : grad_x[n,j] = G@W^T[:, j] + 2*x[n,j] * (G@W^T)[:, d+j]
+//   K2  topk_hw : per (image, prototype) top-T over the spatial axis with
+//                 indices (reference model.py:188-206), lane-per-prototype
+//                 insertion sort, coalesced over P.
+//   K8  argmax_hw: top-1 variant for the push distance argmin
+//                 (reference push.py:134-135).
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//   * wavefront = 64; blocks are 256 threads (4 waves).
+//   * fp32-input MFMA (v_mfma_f32_16x16x4_f32) is exact f32 at the f32
+//     vector rate — the prototype path stays fp32 for parity with the
+//     fp32 oracle while the backbone runs bf16 through MIOpen.
+//   * K (=2d, 128 or 256) is small, so both operand panels are staged in
+//     LDS ONCE per block (no K tiling): A = [x | x^2] rows, B = W columns
+//     stored transposed.  K-stride padded to 2d+4 words so the b128
+//     column-fragment reads are bank-conflict-free (16-lane groups map to
+//     disjoint 4-bank runs).
+//   * MFMA k-order within a 16-wide macro step is permuted (k = 4*kk + i)
+//     so each lane reads its operands with one ds_read_b128; summation
+//     over k commutes so the result is unchanged.
+
+#include <torch/extension.h>
+#include <hip/hip_runtime.h>
+#include <ATen/hip/HIPContext.h>
+
+#define CHECK_IN(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous on device")
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+// ---------------------------------------------------------------------------
+// K1 forward GEMM: out[N,P] = f(bias + [x,x^2] @ W)
+// Template: BM x BN block tile, 4 waves each computing (BM/2)x(BN/2),
+// KMAX = 2d (128 for d=64, 256 for d=128), LDS K-stride = KMAX+4.
+// ---------------------------------------------------------------------------
+
+template <int BM, int BN, int KMAX>
+__global__ __launch_bounds__(256)
+void gmm_fwd_kernel(const float* __restrict__ x,   // [N, d]
+                    const float* __restrict__ w,   // [2d, P]
+                    const float* __restrict__ bias,// [P]
+                    float* __restrict__ out,       // [N, P]
+                    int N, int d, int P, int apply_exp) {
+    constexpr int KS = KMAX + 4;                    // padded LDS k-stride
+    constexpr int WM = BM / 2;                      // per-wave rows
+    constexpr int WN = BN / 2;                      // per-wave cols
+    constexpr int FM = WM / 16;                     // 16x16 frags per wave (rows)
+    constexpr int FN = WN / 16;
+
+    __shared__ float lds[(BM + BN) * KS];
+    float* As = lds;                                // [BM][KS]
+    float* Bs = lds + BM * KS;                      // [BN][KS]
+
+    const int K2 = 2 * d;
+    const int n0 = blockIdx.x * BM;
+    const int p0 = blockIdx.y * BN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+
+    // ---- stage A = [x, x^2] rows n0..n0+BM ------------------------------
+    // each thread loads float4s of x; d % 4 == 0 (checked host-side)
+    {
+        const int vec_per_row = d / 4;
+        for (int t = tid; t < BM * vec_per_row; t += 256) {
+            const int r = t / vec_per_row;
+            const int c4 = t % vec_per_row;
+            const int n = min(n0 + r, N - 1);
+            const float4 v = reinterpret_cast<const float4*>(x + (long)n * d)[c4];
+            float* dst = As + r * KS + c4 * 4;
+            dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
+            float* dst2 = dst + d;
+            dst2[0] = v.x * v.x; dst2[1] = v.y * v.y;
+            dst2[2] = v.z * v.z; dst2[3] = v.w * v.w;
+        }
+    }
+    // ---- stage B transposed: Bs[p - p0][j2] = w[j2][p] ------------------
+    {
+        for (int t = tid; t < K2 * (BN / 4); t += 256) {
+            const int j2 = t / (BN / 4);
+            const int c4 = t % (BN / 4);
+            const int p = p0 + c4 * 4;
+            float4 v;
+            if (p + 3 < P) {
+                v = *reinterpret_cast<const float4*>(w + (long)j2 * P + p);
+            } else {
+                v.x = (p + 0 < P) ? w[(long)j2 * P + p + 0] : 0.f;
+                v.y = (p + 1 < P) ? w[(long)j2 * P + p + 1] : 0.f;
+                v.z = (p + 2 < P) ? w[(long)j2 * P + p + 2] : 0.f;
+                v.w = (p + 3 < P) ? w[(long)j2 * P + p + 3] : 0.f;
+            }
+            Bs[(c4 * 4 + 0) * KS + j2] = v.x;
+            Bs[(c4 * 4 + 1) * KS + j2] = v.y;
+            Bs[(c4 * 4 + 2) * KS + j2] = v.z;
+            Bs[(c4 * 4 + 3) * KS + j2] = v.w;
+        }
+    }
+    __syncthreads();
+
+    // ---- MFMA main loop -------------------------------------------------
+    const int wr = (wave >> 1) * WM;                // wave row offset in tile
+    const int wc = (wave & 1) * WN;                 // wave col offset
+    const int lrow = lane & 15;                     // fragment row/col lane
+    const int kk = lane >> 4;                       // lane k-slot (0..3)
+
+    f32x4 acc[FM][FN];
+    #pragma unroll
+    for (int i = 0; i < FM; ++i)
+        #pragma unroll
+        for (int j = 0; j < FN; ++j)
+            acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    for (int k0 = 0; k0 < K2; k0 += 16) {           // macro K step (16)
+        // per lane: A[row][k0 + 4*kk .. +3], B[col][k0 + 4*kk .. +3]
+        f32x4 afr[FM], bfr[FN];
+        #pragma unroll
+        for (int i = 0; i < FM; ++i) {
+            const float* src = As + (wr + i * 16 + lrow) * KS + k0 + 4 * kk;
+            afr[i] = *reinterpret_cast<const f32x4*>(src);
+        }
+        #pragma unroll
+        for (int j = 0; j < FN; ++j) {
+            const float* src = Bs + (wc + j * 16 + lrow) * KS + k0 + 4 * kk;
+            bfr[j] = *reinterpret_cast<const f32x4*>(src);
+        }
+        #pragma unroll
+        for (int q = 0; q < 4; ++q) {               // permuted k-order, sum commutes
+            #pragma unroll
+            for (int i = 0; i < FM; ++i)
+                #pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                        afr[i][q], bfr[j][q], acc[i][j], 0, 0, 0);
+        }
+    }
+
+    // ---- epilogue: bias + exp, masked store -----------------------------
+    #pragma unroll
+    for (int i = 0; i < FM; ++i) {
+        #pragma unroll
+        for (int j = 0; j < FN; ++j) {
+            const int col = p0 + wc + j * 16 + lrow;
+            if (col >= P) continue;
+            const float b = bias[col];
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = n0 + wr + i * 16 + (kk * 4 + r);
+                if (row >= N) continue;
+                float v = acc[i][j][r] + b;
+                if (apply_exp) v = __expf(v);
+                out[(long)row * P + col] = v;
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K1 backward: gw[N,2d] = G[N,P] @ W^T, then grad_x = gw[:, :d] + 2x*gw[:, d:]
+// K (=P) is large: tiled K loop, BK=32, A=G rows, B=W columns (w[j2][p] with
+// p as K). Output tile = [BM, 2d] (2d <= 256), one block row per BM rows.
+// ---------------------------------------------------------------------------
+
+template <int BM, int KMAX>
+__global__ __launch_bounds__(256)
+void gmm_bwd_kernel(const float* __restrict__ g,   // [N, P]
+                    const float* __restrict__ x,   // [N, d]
+                    const float* __restrict__ w,   // [2d, P]
+                    float* __restrict__ gx,        // [N, d]
+                    int N, int d, int P) {
+    constexpr int BK = 32;
+    constexpr int KS = BK + 4;
+    constexpr int BN = KMAX;                        // all 2d columns
+    constexpr int WM = BM / 2;
+    constexpr int WN = BN / 2;
+    constexpr int FM = WM / 16;
+    constexpr int FN = WN / 16;
+    // LDS serves (a) the staging tiles during the K loop and (b) the full
+    // [BM][KMAX] gw scratch in the epilogue — size for the larger
+    constexpr int LDS_WORDS = ((BM + BN) * KS > BM * KMAX)
+                                  ? (BM + BN) * KS : BM * KMAX;
+
+    __shared__ float lds[LDS_WORDS];
+    float* Gs = lds;                                // [BM][KS]
+    float* Ws = lds + BM * KS;                      // [BN][KS] (transposed)
+
+    const int K2 = 2 * d;
+    const int n0 = blockIdx.x * BM;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wr = (wave >> 1) * WM;
+    const int wc = (wave & 1) * WN;
+    const int lrow = lane & 15;
+    const int kk = lane >> 4;
+
+    f32x4 acc[FM][FN];
+    #pragma unroll
+    for (int i = 0; i < FM; ++i)
+        #pragma unroll
+        for (int j = 0; j < FN; ++j)
+            acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    for (int p0 = 0; p0 < P; p0 += BK) {
+        // stage G tile [BM rows x BK p] : coalesced along p
+        for (int t = tid; t < BM * (BK / 4); t += 256) {
+            const int r = t / (BK / 4);
+            const int c4 = t % (BK / 4);
+            const int n = min(n0 + r, N - 1);
+            const int p = p0 + c4 * 4;
+            float4 v = {0.f, 0.f, 0.f, 0.f};
+            if (p + 3 < P) {
+                v = *reinterpret_cast<const float4*>(g + (long)n * P + p);
+            } else {
+                if (p + 0 < P) v.x = g[(long)n * P + p + 0];
+                if (p + 1 < P) v.y = g[(long)n * P + p + 1];
+                if (p + 2 < P) v.z = g[(long)n * P + p + 2];
+                if (p + 3 < P) v.w = g[(long)n * P + p + 3];
+            }
+            float* dst = Gs + r * KS + c4 * 4;
+            dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
+        }
+        // stage W tile transposed: Ws[j2][pp] = w[j2][p0+pp]
+        for (int t = tid; t < K2 * (BK / 4); t += 256) {
+            const int j2 = t / (BK / 4);
+            const int c4 = t % (BK / 4);
+            const int p = p0 + c4 * 4;
+            float4 v = {0.f, 0.f, 0.f, 0.f};
+            if (p + 3 < P) {
+                v = *reinterpret_cast<const float4*>(w + (long)j2 * P + p);
+            } else {
+                if (p + 0 < P) v.x = w[(long)j2 * P + p + 0];
+                if (p + 1 < P) v.y = w[(long)j2 * P + p + 1];
+                if (p + 2 < P) v.z = w[(long)j2 * P + p + 2];
+                if (p + 3 < P) v.w = w[(long)j2 * P + p + 3];
+            }
+            float* dst = Ws + j2 * KS + c4 * 4;
+            dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
+        }
+        __syncthreads();
+
+        #pragma unroll
+        for (int km = 0; km < BK; km += 16) {
+            f32x4 afr[FM], bfr[FN];
+            #pragma unroll
+            for (int i = 0; i < FM; ++i)
+                afr[i] = *reinterpret_cast<const f32x4*>(
+                    Gs + (wr + i * 16 + lrow) * KS + km + 4 * kk);
+            #pragma unroll
+            for (int j = 0; j < FN; ++j)
+                bfr[j] = *reinterpret_cast<const f32x4*>(
+                    Ws + (wc + j * 16 + lrow) * KS + km + 4 * kk);
+            #pragma unroll
+            for (int q = 0; q < 4; ++q)
+                #pragma unroll
+                for (int i = 0; i < FM; ++i)
+                    #pragma unroll
+                    for (int j = 0; j < FN; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            afr[i][q], bfr[j][q], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    // ---- epilogue: combine the two K halves: gx = gw_lo + 2x*gw_hi ------
+    // acc columns j2 in [0, 2d); column j2 and j2+d pair up.  Each lane owns
+    // col = wc + j*16 + lrow; stage gw into LDS (reuse) then combine.
+    float* gw = lds;                                // reuse as [BM][KMAX]
+    __syncthreads();
+    #pragma unroll
+    for (int i = 0; i < FM; ++i)
+        #pragma unroll
+        for (int j = 0; j < FN; ++j) {
+            const int col = wc + j * 16 + lrow;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = wr + i * 16 + kk * 4 + r;
+                gw[row * KMAX + col] = acc[i][j][r];
+            }
+        }
+    __syncthreads();
+    for (int t = tid; t < BM * d; t += 256) {
+        const int r = t / d;
+        const int c = t % d;
+        const int n = n0 + r;
+        if (n >= N || c >= d) continue;
+        const float xv = x[(long)n * d + c];
+        gx[(long)n * d + c] = gw[r * KMAX + c] + 2.f * xv * gw[r * KMAX + d + c];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K2: top-T over HW per (b, p).  Lane-per-prototype insertion sort.
+// probs [B, HW, P] -> vals [B, P, T], idx [B, P, T] (int32, desc order).
+// Ties: lower hw wins (matches torch.topk / np.argmin-first semantics).
+// ---------------------------------------------------------------------------
+
+template <int TMAX>
+__global__ __launch_bounds__(256)
+void topk_hw_kernel(const float* __restrict__ probs,
+                    float* __restrict__ vals, int* __restrict__ idx,
+                    int B, int HW, int P, int T) {
+    const int p = blockIdx.x * 256 + threadIdx.x;
+    const int b = blockIdx.y;
+    if (p >= P) return;
+
+    float v[TMAX];
+    int ix[TMAX];
+    #pragma unroll
+    for (int t = 0; t < TMAX; ++t) { v[t] = -INFINITY; ix[t] = -1; }
+
+    const float* src = probs + (long)b * HW * P + p;
+    for (int hw = 0; hw < HW; ++hw) {
+        const float val = src[(long)hw * P];
+        if (val > v[T - 1]) {
+            // insertion: strict > keeps earlier hw ahead of equal values
+            int pos = T - 1;
+            while (pos > 0 && val > v[pos - 1]) {
+                v[pos] = v[pos - 1]; ix[pos] = ix[pos - 1]; --pos;
+            }
+            v[pos] = val; ix[pos] = hw;
+        }
+    }
+    float* vdst = vals + ((long)b * P + p) * T;
+    int* idst = idx + ((long)b * P + p) * T;
+    for (int t = 0; t < T; ++t) { vdst[t] = v[t]; idst[t] = ix[t]; }
+}
+
+__global__ __launch_bounds__(256)
+void argmax_hw_kernel(const float* __restrict__ probs,
+                      float* __restrict__ vals, int* __restrict__ idx,
+                      int B, int HW, int P) {
+    const int p = blockIdx.x * 256 + threadIdx.x;
+    const int b = blockIdx.y;
+    if (p >= P) return;
+    const float* src = probs + (long)b * HW * P + p;
+    float best = -INFINITY; int bi = 0;
+    for (int hw = 0; hw < HW; ++hw) {
+        const float val = src[(long)hw * P];
+        if (val > best) { best = val; bi = hw; }
+    }
+    vals[(long)b * P + p] = best;
+    idx[(long)b * P + p] = bi;
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
+
+torch::Tensor gmm_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+                      bool apply_exp) {
+    CHECK_IN(x); CHECK_IN(w); CHECK_IN(bias);
+    TORCH_CHECK(x.dtype() == torch::kFloat32, "gmm_fwd: fp32 only");
+    const int N = x.size(0), d = x.size(1), P = w.size(1);
+    TORCH_CHECK(w.size(0) == 2 * d, "w must be [2d, P]");
+    TORCH_CHECK(d % 8 == 0 && d <= 128, "d must be multiple of 8, <= 128");
+    TORCH_CHECK(P % 4 == 0, "P must be a multiple of 4");
+    auto out = torch::empty({N, P}, x.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    if (d <= 64) {
+        dim3 grid(ceil_div(N, 128), ceil_div(P, 128));
+        hipLaunchKernelGGL((gmm_fwd_kernel<128, 128, 128>), grid, dim3(256), 0,
+                           stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                           bias.data_ptr<float>(), out.data_ptr<float>(),
+                           N, d, P, (int)apply_exp);
+    } else {
+        dim3 grid(ceil_div(N, 64), ceil_div(P, 64));
+        hipLaunchKernelGGL((gmm_fwd_kernel<64, 64, 256>), grid, dim3(256), 0,
+                           stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                           bias.data_ptr<float>(), out.data_ptr<float>(),
+                           N, d, P, (int)apply_exp);
+    }
+    return out;
+}
+
+torch::Tensor gmm_bwd(torch::Tensor g, torch::Tensor x, torch::Tensor w) {
+    CHECK_IN(g); CHECK_IN(x); CHECK_IN(w);
+    const int N = x.size(0), d = x.size(1), P = w.size(1);
+    TORCH_CHECK(g.size(0) == N && g.size(1) == P, "g must be [N, P]");
+    TORCH_CHECK(P % 4 == 0, "P must be a multiple of 4");
+    auto gx = torch::empty({N, d}, x.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    if (d <= 64) {
+        hipLaunchKernelGGL((gmm_bwd_kernel<128, 128>), dim3(ceil_div(N, 128)),
+                           dim3(256), 0, stream,
+                           g.data_ptr<float>(), x.data_ptr<float>(),
+                           w.data_ptr<float>(), gx.data_ptr<float>(), N, d, P);
+    } else {
+        hipLaunchKernelGGL((gmm_bwd_kernel<64, 256>), dim3(ceil_div(N, 64)),
+                           dim3(256), 0, stream,
+                           g.data_ptr<float>(), x.data_ptr<float>(),
+                           w.data_ptr<float>(), gx.data_ptr<float>(), N, d, P);
+    }
+    return gx;
+}
+
+std::vector<torch::Tensor> topk_hw(torch::Tensor probs, long T) {
+    CHECK_IN(probs);
+    const int B = probs.size(0), HW = probs.size(1), P = probs.size(2);
+    TORCH_CHECK(T <= 32, "topk_hw supports T <= 32");
+    TORCH_CHECK(T <= HW, "T must be <= HW");
+    auto vals = torch::empty({B, P, T}, probs.options());
+    auto idx = torch::empty({B, P, T}, probs.options().dtype(torch::kInt32));
+    dim3 grid(ceil_div(P, 256), B);
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL((topk_hw_kernel<32>), grid, dim3(256), 0, stream,
+                       probs.data_ptr<float>(), vals.data_ptr<float>(),
+                       idx.data_ptr<int>(), B, HW, P, (int)T);
+    return {vals, idx};
+}
+
+std::vector<torch::Tensor> argmax_hw(torch::Tensor probs) {
+    CHECK_IN(probs);
+    const int B = probs.size(0), HW = probs.size(1), P = probs.size(2);
+    auto vals = torch::empty({B, P}, probs.options());
+    auto idx = torch::empty({B, P}, probs.options().dtype(torch::kInt32));
+    dim3 grid(ceil_div(P, 256), B);
+    auto stream = at::hip::getCurrentHIPStream();
+    hipLaunchKernelGGL(argmax_hw_kernel, grid, dim3(256), 0, stream,
+                       probs.data_ptr<float>(), vals.data_ptr<float>(),
+                       idx.data_ptr<int>(), B, HW, P);
+    return {vals, idx};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("gmm_fwd", &gmm_fwd, "fused GMM score GEMM forward (MFMA f32)");
+    m.def("gmm_bwd", &gmm_bwd, "GMM score GEMM backward wrt features");
+    m.def("topk_hw", &topk_hw, "per-(b,p) top-T over HW with indices");
+    m.def("argmax_hw", &argmax_hw, "per-(b,p) argmax over HW");
+}

@@ -1,0 +1,121 @@
+"""Bucketed gradient all-reduce overlapped with backward.
+
+Replaces the reference's DataParallel gradient path with
+explicit RCCL collectives: parameters are grouped into flat buckets in
+reverse registration order (≈ backward completion order); when the last
+grad of a bucket lands (post-accumulate-grad hook), the bucket is flattened
+and an async all-reduce is issued, overlapping communication with the rest
+of backward. ``finalize()`` waits for all reduces and scatters averaged
+grads back.
+
+Bucket sizing targets xGMI: each of a GPU's 7 point-to-point links moves
+≈153 GB/s, and RCCL rings are per-link bound, so buckets need to be large
+enough (tens of MB) to amortize ring latency but small enough that the
+first bucket can launch well before backward ends. Default 50 MB
+(~ResNet-50's grads form ~2 buckets).
+"""
+
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+class BucketedGradReducer:
+    def __init__(self, module: torch.nn.Module, comm, bucket_mb: int = 50,
+                 skip_params=()):
+        self.comm = comm
+        self.enabled = comm is not None and comm.is_distributed
+        self.bucket_bytes = bucket_mb * 1024 * 1024
+        self._hooks = []
+        self._works: List = []
+
+        skip = set(id(p) for p in skip_params)
+        modules = module if isinstance(module, (list, tuple)) else [module]
+        params = [p for m in modules for p in m.parameters()
+                  if p.requires_grad and id(p) not in skip]
+        # reverse order ~ backward completion order for sequential nets
+        params = params[::-1]
+
+        self.buckets: List[List[torch.nn.Parameter]] = []
+        cur, cur_bytes = [], 0
+        for p in params:
+            nbytes = p.numel() * p.element_size()
+            if cur and cur_bytes + nbytes > self.bucket_bytes:
+                self.buckets.append(cur)
+                cur, cur_bytes = [], 0
+            cur.append(p)
+            cur_bytes += nbytes
+        if cur:
+            self.buckets.append(cur)
+
+        self._bucket_of = {}
+        self._pending = [0] * len(self.buckets)
+        self._flat: List[Optional[torch.Tensor]] = [None] * len(self.buckets)
+        for bi, bucket in enumerate(self.buckets):
+            for p in bucket:
+                self._bucket_of[id(p)] = bi
+
+        if self.enabled:
+            for p in params:
+                h = p.register_post_accumulate_grad_hook(self._on_grad)
+                self._hooks.append(h)
+
+    # ------------------------------------------------------------------
+    def prepare(self):
+        """Call at the start of every backward."""
+        if not self.enabled:
+            return
+        self._works.clear()
+        for bi, bucket in enumerate(self.buckets):
+            self._pending[bi] = len(bucket)
+
+    def _on_grad(self, p):
+        bi = self._bucket_of[id(p)]
+        self._pending[bi] -= 1
+        if self._pending[bi] == 0:
+            self._launch(bi)
+
+    def _launch(self, bi):
+        bucket = self.buckets[bi]
+        flat = torch._utils._flatten_dense_tensors(
+            [p.grad for p in bucket])
+        flat.div_(self.comm.world_size)
+        work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
+        self._flat[bi] = flat
+        self._works.append((work, bi))
+
+    def finalize(self):
+        """Wait for all reduces and write averaged grads back. Call before
+        optimizer.step()."""
+        if not self.enabled:
+            return
+        # Partially-filled buckets (params frozen mid-run, or unused in this
+        # graph) never auto-launched: reduce their present grads now so no
+        # rank diverges. Ranks must agree on which params got grads.
+        for bi, bucket in enumerate(self.buckets):
+            if 0 < self._pending[bi] < len(bucket):
+                present = [p for p in bucket if p.grad is not None]
+                if present:
+                    flat = torch._utils._flatten_dense_tensors(
+                        [p.grad for p in present])
+                    flat.div_(self.comm.world_size)
+                    dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+                    for p, g in zip(present, torch._utils._unflatten_dense_tensors(
+                            flat, [p.grad for p in present])):
+                        p.grad.copy_(g)
+                self._pending[bi] = 0
+        for work, bi in self._works:
+            work.wait()
+            bucket = self.buckets[bi]
+            synced = torch._utils._unflatten_dense_tensors(
+                self._flat[bi], [p.grad for p in bucket])
+            for p, g in zip(bucket, synced):
+                p.grad.copy_(g)
+            self._flat[bi] = None
+        self._works.clear()
+
+    def remove(self):
+        for h in self._hooks:
+            h.remove()
+        self._hooks.clear()

@@ -39,10 +39,21 @@ class MemoryBank(nn.Module):
         self.mode = mode
         self.fix_length_mult = fix_length_mult
 
-        self.register_buffer('mem', torch.zeros(num_classes, self.cap_cls,
-                                                dim_feature))
+        # flat storage with one extra "trash" class row: rows with the
+        # sentinel label ``num_classes`` land there, which lets the
+        # distributed enqueue push fixed-size padded batches with no host
+        # sync (invalid rows are written to trash and never read)
+        self.register_buffer('_mem_flat',
+                             torch.zeros((num_classes + 1) * self.cap_cls,
+                                         dim_feature))
         self.register_buffer('mem_len', torch.zeros(num_classes, dtype=torch.int64))
         self.register_buffer('head', torch.zeros(num_classes, dtype=torch.int64))
+
+    @property
+    def mem(self) -> torch.Tensor:
+        """[C, cap, d] view of the live storage (excludes the trash row)."""
+        return self._mem_flat[:self.num_classes * self.cap_cls].view(
+            self.num_classes, self.cap_cls, self.dim_feature)
 
     # ------------------------------------------------------------------ push
     @torch.no_grad()
@@ -50,7 +61,9 @@ class MemoryBank(nn.Module):
         """Append (feature, label) pairs FIFO per class. Batched, in-order.
 
         ``feature``: [M, d]; ``label``: [M] int64. Items of one class are
-        inserted in their order of appearance.
+        inserted in their order of appearance. Rows labelled with the
+        sentinel ``num_classes`` are discarded (written to the trash row) —
+        no host sync, used by padded distributed pushes.
         """
         assert feature.dim() == 2 and label.dim() == 1
         assert feature.size(0) == label.size(0)
@@ -60,10 +73,11 @@ class MemoryBank(nn.Module):
         feature = feature.detach()
         label = label.detach()
         cap = self.cap_cls
+        C = self.num_classes
         dev = feature.device
 
         order = torch.argsort(label, stable=True)
-        lab = label[order]
+        lab = label[order]                    # sentinel rows sort last
         fea = feature[order]
 
         # position within each class segment
@@ -74,19 +88,24 @@ class MemoryBank(nn.Module):
         seg_first = idx[change]                                     # [#segments]
         within = idx - seg_first[seg_id]                            # [M]
 
-        counts = torch.zeros(self.num_classes, dtype=torch.int64, device=dev)
+        counts = torch.zeros(C + 1, dtype=torch.int64, device=dev)
         counts.scatter_add_(0, lab, torch.ones_like(lab))
 
-        # if a class pushes more than cap at once, keep only the newest cap
-        # (positions keep their original within-offsets so the advanced head
-        # still points at the oldest surviving row)
+        # if a class pushes more than cap at once, write only the newest cap
+        # (stale rows' flat indices are redirected to the trash row so the
+        # whole batch scatters in one op; the advanced head still points at
+        # the oldest surviving row)
         keep = within >= (counts[lab] - cap)
-        lab, fea, within = lab[keep], fea[keep], within[keep]
+        is_trash = (lab == C) | ~keep
+        pos = (self.head[torch.clamp(lab, max=C - 1)] + within) % cap
+        flat = torch.where(is_trash,
+                           C * cap + (within % cap),
+                           torch.clamp(lab, max=C - 1) * cap + pos)
+        self._mem_flat[flat] = fea.to(self._mem_flat.dtype)
 
-        pos = (self.head[lab] + within) % cap
-        self.mem[lab, pos] = fea.to(self.mem.dtype)
-        self.head = (self.head + counts) % cap
-        self.mem_len = torch.minimum(self.mem_len + counts,
+        counts_real = counts[:C]
+        self.head = (self.head + counts_real) % cap
+        self.mem_len = torch.minimum(self.mem_len + counts_real,
                                      torch.full_like(self.mem_len, cap))
 
     # ------------------------------------------------------------------ pull
