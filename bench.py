@@ -54,7 +54,7 @@ def main():
     world = int(os.environ.get('WORLD_SIZE', '1'))
     comm = Comm() if world > 1 else None
     device = comm.device if comm is not None else (
-        torch.device('cuda') if torch.cuda.is_available() else torch.device('cpu'))
+        torch.device('cuda', 0) if torch.cuda.is_available() else torch.device('cpu'))
     if device.type == 'cuda':
         torch.cuda.set_device(device)
     rank = comm.rank if comm is not None else 0
