@@ -228,13 +228,22 @@ def em_e_step(x: torch.Tensor, means: torch.Tensor, covs: torch.Tensor,
     ``pi``: [G, K]. Returns (weighted_log_prob [G, N, K], log_resp [G, N, K]).
     Reference: model.py:303-336 (`_e_step` + `_estimate_log_prob`, run per
     class in a Python loop there; batched over classes here).
+
+    Uses the same quadratic-expansion GEMM form as K1 (gmm_expand_params):
+    a [G,N,2d]x[G,2d,K] bmm instead of the [G,N,K,d] broadcast temporary
+    (which is ~4 GB at the default shapes and was the EM hot spot).
     """
     d = x.shape[-1]
     sig = covs + eps                                          # [G, K, d]
-    diff = x.unsqueeze(2) - means.unsqueeze(1)                # [G, N, K, d]
-    log_p = (diff / sig.unsqueeze(1)).pow(2).sum(-1)          # [G, N, K]
-    log_sigma = torch.log(sig).sum(-1)                        # [G, K]
-    log_prob = -0.5 * d * LOG_2PI - log_sigma.unsqueeze(1) - 0.5 * log_p
+    inv_var = 1.0 / (sig * sig)
+    A = (means * inv_var).transpose(1, 2)                     # [G, d, K]
+    Bq = (-0.5 * inv_var).transpose(1, 2)                     # [G, d, K]
+    bias = (-0.5 * d * LOG_2PI
+            - torch.log(sig).sum(-1)
+            - 0.5 * (means * means * inv_var).sum(-1))        # [G, K]
+    x2 = torch.cat([x, x * x], dim=2)                         # [G, N, 2d]
+    W = torch.cat([A, Bq], dim=1)                             # [G, 2d, K]
+    log_prob = torch.baddbmm(bias.unsqueeze(1), x2, W)        # [G, N, K]
     wlp = log_prob + torch.log(pi + eps).unsqueeze(1)         # [G, N, K]
     log_norm = torch.logsumexp(wlp, dim=2, keepdim=True)
     return wlp, wlp - log_norm
