@@ -51,6 +51,11 @@ def main():
     from mgproto_amd.parallel import Comm, BucketedGradReducer, make_dp_correct
     from mgproto_amd.engine.trainer import EMRunner
 
+    # MIOpen exhaustive find for the (static) conv shapes: the search cost
+    # lands in warmup; without it MIOpen's fallback solvers dominate the step
+    # (conv bwd-weight measured at 74% of GPU time under FIND_MODE=FAST)
+    torch.backends.cudnn.benchmark = True
+
     world = int(os.environ.get('WORLD_SIZE', '1'))
     comm = Comm() if world > 1 else None
     device = comm.device if comm is not None else (

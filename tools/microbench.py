@@ -3,7 +3,11 @@
 vs losses vs optimizer. GPU-only diagnostic tool."""
 
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
@@ -30,8 +34,10 @@ def main():
     ap.add_argument('--addon', type=str, default='regular_upsample')
     ap.add_argument('--no-channels-last', action='store_true')
     ap.add_argument('--no-amp', action='store_true')
+    ap.add_argument('--no-find', action='store_true')
     args = ap.parse_args()
 
+    torch.backends.cudnn.benchmark = not args.no_find
     dev = torch.device('cuda', 0)
     C, K, d = 200, 10, 64
     torch.manual_seed(0)
