@@ -313,56 +313,63 @@ class MGProto(nn.Module):
         """Batched EM over all dirty+full classes (reference model.py:277-301).
 
         E-step + closed-form diversified M-step (ops.em_e_step /
-        ops.em_m_step_grads), one fused launch set per EM loop for ALL dirty
-        classes instead of a per-class Python loop; per-class Adam on the
-        means; pi momentum tau; priors written into the last layer.
+        ops.em_m_step_grads), one fused launch set per EM loop for ALL
+        classes at once instead of a per-class Python loop; per-class Adam
+        on the means; pi momentum tau; priors written into the last layer.
+
+        Shapes are STATIC ([C, cap, d] with an active-class mask) and there
+        is no host synchronization, so the whole update runs asynchronously
+        on a side HIP stream (EMRunner) and is hipGraph-capturable; inactive
+        classes are computed and discarded (in steady state, with
+        update_interval=1, every class in the batch history is dirty anyway).
         """
         dirty = self.memory_updated_cls.clone()
         self.memory_updated_cls.zero_()
-        active = dirty & self.queue.full_mask()
-        classes = active.nonzero(as_tuple=False).flatten()
-        if classes.numel() == 0:
-            return
+        active = (dirty & self.queue.full_mask()).view(-1, 1)    # [C, 1]
+        if not self.queue.mem.is_cuda and not bool(active.any()):
+            return  # CPU path: skip the (cheap) masked compute entirely
 
         C, K = self.num_classes, self.num_prototypes_per_class
-        x = self.queue.pull_dense(classes)                      # [G, cap, d]
+        x = self.queue.mem                                       # [C, cap, d]
         N = x.shape[1]
         w = self.last_layer.weight.data                          # [C, P]
-        own = w.view(C, C, K)[torch.arange(C, device=w.device),
-                              torch.arange(C, device=w.device)]  # [C, K]
-        pi_old = own[classes]                                    # [G, K]
-        means = self.prototype_means.data[classes]               # [G, K, d]
-        covs = self.prototype_covs.data[classes]
+        diag = torch.arange(C, device=w.device)
+        pi_cur = w.view(C, C, K)[diag, diag]                     # [C, K]
+        pi_old = pi_cur.clone()
+        means = self.prototype_means.data.clone()                # [C, K, d]
+        covs = self.prototype_covs.data
 
         for _ in range(self.num_em_loop):
             wlp, log_resp = ops.em_e_step(x, means, covs, pi_old)
             grad, pi_unnorm = ops.em_m_step_grads(
                 x, log_resp, wlp, means, covs, alpha=self.alpha,
                 lamda=self.lamda)
-            means = self._em_adam_step(classes, means, grad)
+            means = self._em_adam_step(active, means, grad)
             pi = pi_unnorm / N
             pi_old = momentum_update(pi_old, pi, self.tau)
 
-        self.prototype_means.data[classes] = means
-        # scatter pi back into the class-masked last layer
-        pcols = (classes.unsqueeze(1) * K
-                 + torch.arange(K, device=w.device).unsqueeze(0))  # [G, K]
-        w[classes.unsqueeze(1), pcols] = pi_old
+        act3 = active.unsqueeze(-1)                              # [C, 1, 1]
+        self.prototype_means.data.copy_(
+            torch.where(act3, means, self.prototype_means.data))
+        w.view(C, C, K)[diag, diag] = torch.where(active, pi_old, pi_cur)
 
-    def _em_adam_step(self, classes, means, grad):
+    def _em_adam_step(self, active, means, grad):
+        """Masked per-class Adam: only active classes advance state/means."""
         b1, b2 = self.adam_betas
-        self._em_step[classes] += 1
-        step = self._em_step[classes].to(means.dtype).view(-1, 1, 1)
-        m = b1 * self._em_exp_avg[classes] + (1 - b1) * grad
-        v = b2 * self._em_exp_avg_sq[classes] + (1 - b2) * grad * grad
-        self._em_exp_avg[classes] = m
-        self._em_exp_avg_sq[classes] = v
+        act3 = active.unsqueeze(-1)                              # [C, 1, 1]
+        self._em_step += active.view(-1).to(self._em_step.dtype)
+        step = self._em_step.to(means.dtype).view(-1, 1, 1)
+        m = b1 * self._em_exp_avg + (1 - b1) * grad
+        v = b2 * self._em_exp_avg_sq + (1 - b2) * grad * grad
+        self._em_exp_avg.copy_(torch.where(act3, m, self._em_exp_avg))
+        self._em_exp_avg_sq.copy_(torch.where(act3, v, self._em_exp_avg_sq))
         bc1 = 1 - torch.pow(torch.tensor(b1, dtype=means.dtype,
                                          device=means.device), step)
         bc2 = 1 - torch.pow(torch.tensor(b2, dtype=means.dtype,
                                          device=means.device), step)
         denom = (v / bc2).sqrt() + self.adam_eps
-        return means - self.prototype_lr * (m / bc1) / denom
+        stepped = means - self.prototype_lr * (m / bc1) / denom
+        return torch.where(act3, stepped, means)
 
     # --------------------------------------------------------------- scoring
     def _score(self, x, mu, var, pi, as_average=True, eps=1e-10):

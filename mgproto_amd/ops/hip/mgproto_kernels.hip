@@ -302,12 +302,30 @@ void gmm_bwd_kernel(const float* __restrict__ g,   // [N, P]
 // ---------------------------------------------------------------------------
 
 template <int TMAX>
+__device__ __forceinline__
+void topk_insert(float val, int hw, float (&v)[TMAX], int (&ix)[TMAX], int T) {
+    // insertion: strict > keeps earlier hw ahead of equal values
+    if (val > v[T - 1]) {
+        int pos = T - 1;
+        while (pos > 0 && val > v[pos - 1]) {
+            v[pos] = v[pos - 1]; ix[pos] = ix[pos - 1]; --pos;
+        }
+        v[pos] = val; ix[pos] = hw;
+    }
+}
+
+// Phase 1: each block scans one HW chunk for 256 prototypes of one image,
+// keeping a per-lane top-T; hw loop unrolled x4 to keep loads in flight.
+// nc == 1 writes the final result directly; otherwise partials go to
+// scratch [B, P, nc, T].
+template <int TMAX>
 __global__ __launch_bounds__(256)
 void topk_hw_kernel(const float* __restrict__ probs,
                     float* __restrict__ vals, int* __restrict__ idx,
-                    int B, int HW, int P, int T) {
+                    int B, int HW, int P, int T, int nc, int chunk) {
     const int p = blockIdx.x * 256 + threadIdx.x;
     const int b = blockIdx.y;
+    const int c = blockIdx.z;
     if (p >= P) return;
 
     float v[TMAX];
@@ -315,21 +333,63 @@ void topk_hw_kernel(const float* __restrict__ probs,
     #pragma unroll
     for (int t = 0; t < TMAX; ++t) { v[t] = -INFINITY; ix[t] = -1; }
 
+    const int hw0 = c * chunk;
+    const int hw1 = min(hw0 + chunk, HW);
     const float* src = probs + (long)b * HW * P + p;
-    for (int hw = 0; hw < HW; ++hw) {
-        const float val = src[(long)hw * P];
-        if (val > v[T - 1]) {
-            // insertion: strict > keeps earlier hw ahead of equal values
-            int pos = T - 1;
-            while (pos > 0 && val > v[pos - 1]) {
-                v[pos] = v[pos - 1]; ix[pos] = ix[pos - 1]; --pos;
-            }
-            v[pos] = val; ix[pos] = hw;
-        }
+    int hw = hw0;
+    for (; hw + 4 <= hw1; hw += 4) {
+        const float a0 = src[(long)(hw + 0) * P];
+        const float a1 = src[(long)(hw + 1) * P];
+        const float a2 = src[(long)(hw + 2) * P];
+        const float a3 = src[(long)(hw + 3) * P];
+        topk_insert(a0, hw + 0, v, ix, T);
+        topk_insert(a1, hw + 1, v, ix, T);
+        topk_insert(a2, hw + 2, v, ix, T);
+        topk_insert(a3, hw + 3, v, ix, T);
     }
+    for (; hw < hw1; ++hw) topk_insert(src[(long)hw * P], hw, v, ix, T);
+
+    float* vdst;
+    int* idst;
+    if (nc == 1) {
+        vdst = vals + ((long)b * P + p) * T;
+        idst = idx + ((long)b * P + p) * T;
+    } else {
+        vdst = vals + (((long)b * P + p) * nc + c) * T;
+        idst = idx + (((long)b * P + p) * nc + c) * T;
+    }
+    for (int t = 0; t < T; ++t) { vdst[t] = v[t]; idst[t] = ix[t]; }
+}
+
+// Phase 2: merge nc sorted partial lists per (b, p) into the final top-T.
+// Chunks are visited in order with strict >, so the lowest hw wins ties.
+__global__ __launch_bounds__(256)
+void topk_merge_kernel(const float* __restrict__ pvals,
+                       const int* __restrict__ pidx,
+                       float* __restrict__ vals, int* __restrict__ idx,
+                       int B, int P, int T, int nc) {
+    const int p = blockIdx.x * 256 + threadIdx.x;
+    const int b = blockIdx.y;
+    if (p >= P) return;
+    const float* vsrc = pvals + ((long)b * P + p) * nc * T;
+    const int* isrc = pidx + ((long)b * P + p) * nc * T;
+    int head[8];
+    for (int c = 0; c < nc; ++c) head[c] = 0;
     float* vdst = vals + ((long)b * P + p) * T;
     int* idst = idx + ((long)b * P + p) * T;
-    for (int t = 0; t < T; ++t) { vdst[t] = v[t]; idst[t] = ix[t]; }
+    for (int t = 0; t < T; ++t) {
+        float best = -INFINITY;
+        int bc = 0;
+        for (int c = 0; c < nc; ++c) {
+            if (head[c] < T) {
+                const float hv = vsrc[c * T + head[c]];
+                if (hv > best) { best = hv; bc = c; }
+            }
+        }
+        vdst[t] = best;
+        idst[t] = (best == -INFINITY) ? -1 : isrc[bc * T + head[bc]];
+        head[bc]++;
+    }
 }
 
 __global__ __launch_bounds__(256)
@@ -409,11 +469,34 @@ std::vector<torch::Tensor> topk_hw(torch::Tensor probs, long T) {
     TORCH_CHECK(T <= HW, "T must be <= HW");
     auto vals = torch::empty({B, P, T}, probs.options());
     auto idx = torch::empty({B, P, T}, probs.options().dtype(torch::kInt32));
-    dim3 grid(ceil_div(P, 256), B);
     auto stream = at::hip::getCurrentHIPStream();
-    hipLaunchKernelGGL((topk_hw_kernel<32>), grid, dim3(256), 0, stream,
-                       probs.data_ptr<float>(), vals.data_ptr<float>(),
-                       idx.data_ptr<int>(), B, HW, P, (int)T);
+
+    // chunk HW so the grid comfortably oversubscribes the 256 CUs
+    const long base_blocks = (long)ceil_div(P, 256) * B;
+    int nc = 1;
+    while (nc < 8 && base_blocks * nc < 2048 && (HW / (nc * 2)) >= (int)T
+           && HW / nc > 64)
+        nc *= 2;
+    const int chunk = ceil_div(HW, nc);
+
+    dim3 grid(ceil_div(P, 256), B, nc);
+    if (nc == 1) {
+        hipLaunchKernelGGL((topk_hw_kernel<32>), grid, dim3(256), 0, stream,
+                           probs.data_ptr<float>(), vals.data_ptr<float>(),
+                           idx.data_ptr<int>(), B, HW, P, (int)T, 1, chunk);
+    } else {
+        auto pv = torch::empty({B, P, nc, (int)T}, probs.options());
+        auto pi = torch::empty({B, P, nc, (int)T},
+                               probs.options().dtype(torch::kInt32));
+        hipLaunchKernelGGL((topk_hw_kernel<32>), grid, dim3(256), 0, stream,
+                           probs.data_ptr<float>(), pv.data_ptr<float>(),
+                           pi.data_ptr<int>(), B, HW, P, (int)T, nc, chunk);
+        hipLaunchKernelGGL(topk_merge_kernel, dim3(ceil_div(P, 256), B),
+                           dim3(256), 0, stream,
+                           pv.data_ptr<float>(), pi.data_ptr<int>(),
+                           vals.data_ptr<float>(), idx.data_ptr<int>(),
+                           B, P, (int)T, nc);
+    }
     return {vals, idx};
 }
 
