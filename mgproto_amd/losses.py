@@ -45,8 +45,9 @@ class Proxy_Anchor(nn.Module):
         pos_exp = torch.exp(-self.beta * (cos - self.mrg))
         neg_exp = torch.exp(self.beta * (cos + self.mrg))
 
-        with_pos_proxies = torch.nonzero(P_one_hot.sum(dim=0) != 0).squeeze(dim=1)
-        num_valid_proxies = max(len(with_pos_proxies), 1)
+        # device-side count of classes present in the batch (the reference's
+        # torch.nonzero + len() is a host sync and blocks hipGraph capture)
+        num_valid_proxies = (P_one_hot.sum(dim=0) != 0).sum().clamp(min=1)
 
         P_sim_sum = torch.where(P_one_hot == 1, pos_exp,
                                 torch.zeros_like(pos_exp)).sum(dim=0)
