@@ -303,14 +303,21 @@ void gmm_bwd_kernel(const float* __restrict__ g,   // [N, P]
 
 template <int TMAX>
 __device__ __forceinline__
-void topk_insert(float val, int hw, float (&v)[TMAX], int (&ix)[TMAX], int T) {
-    // insertion: strict > keeps earlier hw ahead of equal values
-    if (val > v[T - 1]) {
-        int pos = T - 1;
-        while (pos > 0 && val > v[pos - 1]) {
-            v[pos] = v[pos - 1]; ix[pos] = ix[pos - 1]; --pos;
+void topk_insert(float val, int hw, float (&v)[TMAX], int (&ix)[TMAX]) {
+    // Fully-unrolled predicated insertion: ALL array indexing is
+    // compile-time static so v/ix stay in VGPRs (a dynamic-index shift
+    // loop sends them to scratch — measured 8x slower).  Strict compares
+    // keep earlier hw ahead of equal values. The guard branch makes the
+    // shift cost appear only on actual inserts (~T*ln(HW/T) per row).
+    if (val > v[TMAX - 1]) {
+        #pragma unroll
+        for (int t = TMAX - 1; t >= 1; --t) {
+            const bool shift = val > v[t - 1];
+            const bool land = (val > v[t]) && !shift;
+            v[t] = shift ? v[t - 1] : (land ? val : v[t]);
+            ix[t] = shift ? ix[t - 1] : (land ? hw : ix[t]);
         }
-        v[pos] = val; ix[pos] = hw;
+        if (val > v[0]) { v[0] = val; ix[0] = hw; }
     }
 }
 
@@ -342,12 +349,12 @@ void topk_hw_kernel(const float* __restrict__ probs,
         const float a1 = src[(long)(hw + 1) * P];
         const float a2 = src[(long)(hw + 2) * P];
         const float a3 = src[(long)(hw + 3) * P];
-        topk_insert(a0, hw + 0, v, ix, T);
-        topk_insert(a1, hw + 1, v, ix, T);
-        topk_insert(a2, hw + 2, v, ix, T);
-        topk_insert(a3, hw + 3, v, ix, T);
+        topk_insert(a0, hw + 0, v, ix);
+        topk_insert(a1, hw + 1, v, ix);
+        topk_insert(a2, hw + 2, v, ix);
+        topk_insert(a3, hw + 3, v, ix);
     }
-    for (; hw < hw1; ++hw) topk_insert(src[(long)hw * P], hw, v, ix, T);
+    for (; hw < hw1; ++hw) topk_insert(src[(long)hw * P], hw, v, ix);
 
     float* vdst;
     int* idst;
@@ -374,21 +381,24 @@ void topk_merge_kernel(const float* __restrict__ pvals,
     const float* vsrc = pvals + ((long)b * P + p) * nc * T;
     const int* isrc = pidx + ((long)b * P + p) * nc * T;
     int head[8];
-    for (int c = 0; c < nc; ++c) head[c] = 0;
+    #pragma unroll
+    for (int c = 0; c < 8; ++c) head[c] = 0;
     float* vdst = vals + ((long)b * P + p) * T;
     int* idst = idx + ((long)b * P + p) * T;
     for (int t = 0; t < T; ++t) {
         float best = -INFINITY;
-        int bc = 0;
-        for (int c = 0; c < nc; ++c) {
-            if (head[c] < T) {
+        int bc = 0, bpos = 0;
+        #pragma unroll
+        for (int c = 0; c < 8; ++c) {            // static indexing only
+            if (c < nc && head[c] < T) {
                 const float hv = vsrc[c * T + head[c]];
-                if (hv > best) { best = hv; bc = c; }
+                if (hv > best) { best = hv; bc = c; bpos = c * T + head[c]; }
             }
         }
         vdst[t] = best;
-        idst[t] = (best == -INFINITY) ? -1 : isrc[bc * T + head[bc]];
-        head[bc]++;
+        idst[t] = (best == -INFINITY) ? -1 : isrc[bpos];
+        #pragma unroll
+        for (int c = 0; c < 8; ++c) head[c] += (c == bc) ? 1 : 0;
     }
 }
 
@@ -480,17 +490,28 @@ std::vector<torch::Tensor> topk_hw(torch::Tensor probs, long T) {
     const int chunk = ceil_div(HW, nc);
 
     dim3 grid(ceil_div(P, 256), B, nc);
+    auto launch_phase1 = [&](float* vp, int* ip) {
+        // smallest register-list instantiation that fits T
+        if (T <= 8)
+            hipLaunchKernelGGL((topk_hw_kernel<8>), grid, dim3(256), 0, stream,
+                               probs.data_ptr<float>(), vp, ip,
+                               B, HW, P, (int)T, nc, chunk);
+        else if (T <= 20)
+            hipLaunchKernelGGL((topk_hw_kernel<20>), grid, dim3(256), 0, stream,
+                               probs.data_ptr<float>(), vp, ip,
+                               B, HW, P, (int)T, nc, chunk);
+        else
+            hipLaunchKernelGGL((topk_hw_kernel<32>), grid, dim3(256), 0, stream,
+                               probs.data_ptr<float>(), vp, ip,
+                               B, HW, P, (int)T, nc, chunk);
+    };
     if (nc == 1) {
-        hipLaunchKernelGGL((topk_hw_kernel<32>), grid, dim3(256), 0, stream,
-                           probs.data_ptr<float>(), vals.data_ptr<float>(),
-                           idx.data_ptr<int>(), B, HW, P, (int)T, 1, chunk);
+        launch_phase1(vals.data_ptr<float>(), idx.data_ptr<int>());
     } else {
         auto pv = torch::empty({B, P, nc, (int)T}, probs.options());
         auto pi = torch::empty({B, P, nc, (int)T},
                                probs.options().dtype(torch::kInt32));
-        hipLaunchKernelGGL((topk_hw_kernel<32>), grid, dim3(256), 0, stream,
-                           probs.data_ptr<float>(), pv.data_ptr<float>(),
-                           pi.data_ptr<int>(), B, HW, P, (int)T, nc, chunk);
+        launch_phase1(pv.data_ptr<float>(), pi.data_ptr<int>());
         hipLaunchKernelGGL(topk_merge_kernel, dim3(ceil_div(P, 256), B),
                            dim3(256), 0, stream,
                            pv.data_ptr<float>(), pi.data_ptr<int>(),
