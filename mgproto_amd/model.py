@@ -363,10 +363,10 @@ class MGProto(nn.Module):
         v = b2 * self._em_exp_avg_sq + (1 - b2) * grad * grad
         self._em_exp_avg.copy_(torch.where(act3, m, self._em_exp_avg))
         self._em_exp_avg_sq.copy_(torch.where(act3, v, self._em_exp_avg_sq))
-        bc1 = 1 - torch.pow(torch.tensor(b1, dtype=means.dtype,
-                                         device=means.device), step)
-        bc2 = 1 - torch.pow(torch.tensor(b2, dtype=means.dtype,
-                                         device=means.device), step)
+        # scalar**tensor stays on device (a torch.tensor(...) here would be a
+        # pageable H2D copy and break hipGraph capture)
+        bc1 = 1 - torch.pow(b1, step)
+        bc2 = 1 - torch.pow(b2, step)
         denom = (v / bc2).sqrt() + self.adam_eps
         stepped = means - self.prototype_lr * (m / bc1) / denom
         return torch.where(act3, stepped, means)

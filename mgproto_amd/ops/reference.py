@@ -169,7 +169,8 @@ def mixture_head(vals: torch.Tensor, weight: torch.Tensor, C: int, K: int
     Reference: model.py:222 (a Python loop of F.linear over T levels).
     """
     B, P, T = vals.shape
-    pi = weight.view(C, C, K)[torch.arange(C), torch.arange(C)]   # [C, K]
+    diag = torch.arange(C, device=weight.device)  # device-side: capture-safe
+    pi = weight.view(C, C, K)[diag, diag]                          # [C, K]
     v = vals.view(B, C, K, T)
     return torch.einsum('bckt,ck->bct', v, pi)
 
