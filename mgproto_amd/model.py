@@ -196,8 +196,11 @@ class MGProto(nn.Module):
         self.capacity_pc = mem_capacity
         self.queue = MemoryBank(self.num_classes, d,
                                 self.capacity_pc * self.num_classes, mode='all')
-        self.register_buffer('memory_updated_cls',
-                             torch.zeros(self.num_classes, dtype=torch.bool),
+        # dirty-class flags with one extra sentinel slot so fixed-size
+        # enqueue batches (labels == C for padding) index it without any
+        # masked compaction (hipGraph-capturable)
+        self.register_buffer('_memory_updated',
+                             torch.zeros(self.num_classes + 1, dtype=torch.bool),
                              persistent=False)
         self.iteration_counter = nn.Parameter(torch.zeros(1), requires_grad=False)
 
@@ -273,9 +276,14 @@ class MGProto(nn.Module):
 
         return torch.log(final_probs), x_auxiliary
 
+    @property
+    def memory_updated_cls(self):
+        """[C] dirty flags view (reference model.py:167 surface)."""
+        return self._memory_updated[:self.num_classes]
+
     def _local_enqueue(self, feats, labels):
         self.queue.push(feats, labels)
-        self.memory_updated_cls[labels] = True
+        self._memory_updated[labels] = True
 
     # --------------------------------------------------- log-prob API parity
     def compute_log_prob(self, _fea, n_block: int = 4, c_block: int = 1,
@@ -323,8 +331,8 @@ class MGProto(nn.Module):
         classes are computed and discarded (in steady state, with
         update_interval=1, every class in the batch history is dirty anyway).
         """
-        dirty = self.memory_updated_cls.clone()
-        self.memory_updated_cls.zero_()
+        dirty = self._memory_updated[:self.num_classes].clone()
+        self._memory_updated.zero_()
         active = (dirty & self.queue.full_mask()).view(-1, 1)    # [C, 1]
         if not self.queue.mem.is_cuda and not bool(active.any()):
             return  # CPU path: skip the (cheap) masked compute entirely

@@ -192,7 +192,9 @@ def enqueue_candidates(feat: torch.Tensor, top1_idx: torch.Tensor,
     samples in batch order, patch indices via torch.unique [sorted]).
 
     The reference's per-sample double Python loop (model.py:240-245) is the
-    K5 hot spot; this is one sort + mask, no host sync.
+    K5 hot spot; this is one sort + mask with STATIC [B*K] output shape and
+    no host sync (hipGraph-capturable): duplicate rows carry the sentinel
+    label C, which MemoryBank.push discards device-side.
     """
     B = gt.shape[0]
     device = feat.device
@@ -208,11 +210,10 @@ def enqueue_candidates(feat: torch.Tensor, top1_idx: torch.Tensor,
     s_idx = s_idx[order]
     first = first[order]
     labels = gt[order].unsqueeze(1).expand(B, K)
-    sel = first.reshape(-1)
-    hw = s_idx.reshape(-1)[sel]                               # [M]
-    lab = labels.reshape(-1)[sel]                             # [M]
-    b_of = order.unsqueeze(1).expand(B, K).reshape(-1)[sel]   # [M]
-    rows = b_of * HW + hw
+    lab = torch.where(first, labels,
+                      torch.full_like(labels, C)).reshape(-1)  # [B*K]
+    b_of = order.unsqueeze(1).expand(B, K).reshape(-1)
+    rows = b_of * HW + s_idx.reshape(-1)
     return feat.index_select(0, rows), lab
 
 
