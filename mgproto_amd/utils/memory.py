@@ -80,13 +80,14 @@ class MemoryBank(nn.Module):
         lab = label[order]                    # sentinel rows sort last
         fea = feature[order]
 
-        # position within each class segment
+        # position within each class segment (cummax of segment starts —
+        # boolean compaction would be a host sync and break graph capture)
         change = torch.ones(M, dtype=torch.bool, device=dev)
         change[1:] = lab[1:] != lab[:-1]
-        seg_id = torch.cumsum(change.long(), 0) - 1                 # [M]
         idx = torch.arange(M, device=dev)
-        seg_first = idx[change]                                     # [#segments]
-        within = idx - seg_first[seg_id]                            # [M]
+        seg_start = torch.cummax(torch.where(change, idx,
+                                             torch.zeros_like(idx)), 0)[0]
+        within = idx - seg_start                                    # [M]
 
         counts = torch.zeros(C + 1, dtype=torch.int64, device=dev)
         counts.scatter_add_(0, lab, torch.ones_like(lab))
