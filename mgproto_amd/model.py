@@ -283,7 +283,9 @@ class MGProto(nn.Module):
 
     def _local_enqueue(self, feats, labels):
         self.queue.push(feats, labels)
-        self._memory_updated[labels] = True
+        # index_fill_ keeps the scalar on device (index_put_ with a Python
+        # bool uploads a host tensor and breaks graph capture)
+        self._memory_updated.index_fill_(0, labels, True)
 
     # --------------------------------------------------- log-prob API parity
     def compute_log_prob(self, _fea, n_block: int = 4, c_block: int = 1,

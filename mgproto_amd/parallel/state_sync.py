@@ -52,7 +52,7 @@ class DistributedEnqueue:
         all_feats = flat[:, :-1]
         all_labels = flat[:, -1].to(torch.int64)
         self.model.queue.push(all_feats, all_labels)
-        self.model._memory_updated[all_labels] = True
+        self.model._memory_updated.index_fill_(0, all_labels, True)
 
 
 def make_dp_correct(model, comm: Comm, train_batch_size: int):
