@@ -1,0 +1,82 @@
+"""DataLoader construction for the real-image pipeline.
+
+Reproduces the reference's five loaders (train / push / test / ood1 / ood2,
+main.py:96-163) with the same transform stacks, but with worker processes
+and distributed sharding (the reference ran num_workers=0 on the main
+thread — SURVEY.md hard part #5)."""
+
+from typing import Tuple
+
+import torch
+from torch.utils.data import DataLoader, Subset
+
+from . import transforms as T
+from .folder import ImageFolder
+from .preprocess import mean, std
+
+
+def _collate(batch):
+    return (torch.stack([b[0] for b in batch]),
+            torch.tensor([b[1] for b in batch]),
+            torch.tensor([b[2] for b in batch]))
+
+
+def _shard(ds, world, rank):
+    if world > 1:
+        return Subset(ds, list(range(rank, len(ds), world)))
+    return ds
+
+
+def build_image_loaders(cfg, world: int = 1, rank: int = 0) -> Tuple:
+    img_size = cfg.img_size
+    normalize = T.Normalize(mean=mean, std=std)
+
+    train_tf = T.Compose([
+        T.RandomPerspective(distortion_scale=0.2, p=0.5),
+        T.ColorJitter((0.6, 1.4), (0.6, 1.4), (0.6, 1.4), (-0.02, 0.02)),
+        T.RandomHorizontalFlip(),
+        T.RandomAffine(degrees=25, shear=(-15, 15), translate=[0.05, 0.05]),
+        T.RandomResizedCrop(size=(img_size, img_size), scale=(0.60, 1.0)),
+        T.ToTensor(),
+        normalize,
+    ])
+    push_tf = T.Compose([
+        T.Resize(size=(img_size, img_size)),
+        T.ToTensor(),
+    ])
+    test_tf = T.Compose([
+        T.Resize(img_size + 32),
+        T.CenterCrop(img_size),
+        T.ToTensor(),
+        normalize,
+    ])
+    ood_tf = T.Compose([
+        T.Resize(size=(img_size, img_size)),
+        T.ToTensor(),
+        normalize,
+    ])
+
+    train_ds = ImageFolder(cfg.train_dir, train_tf)
+    push_ds = ImageFolder(cfg.train_push_dir, push_tf)
+    test_ds = ImageFolder(cfg.test_dir, test_tf)
+
+    def mk(ds, bs, shuffle=False):
+        return DataLoader(ds, batch_size=bs, shuffle=shuffle,
+                          num_workers=cfg.num_workers, pin_memory=True,
+                          collate_fn=_collate, persistent_workers=cfg.num_workers > 0)
+
+    train_loader = mk(_shard(train_ds, world, rank), cfg.train_batch_size,
+                      shuffle=True)
+    push_loader = mk(_shard(push_ds, world, rank), cfg.train_push_batch_size)
+    test_loader = mk(_shard(test_ds, world, rank), cfg.test_batch_size)
+
+    import os
+    ood_loaders = []
+    for d in (cfg.test_dir_ood1, cfg.test_dir_ood2):
+        if os.path.isdir(d):
+            ood_loaders.append(mk(_shard(ImageFolder(d, ood_tf), world, rank),
+                                  cfg.test_batch_size))
+        else:
+            ood_loaders.append(None)
+    return (train_loader, push_loader, test_loader,
+            ood_loaders[0], ood_loaders[1])

@@ -76,6 +76,10 @@ def push_prototypes(dataloader,
     P = model.num_prototypes
     K = model.num_prototypes_per_class
     dataset = dataloader.dataset
+    # pass-2 re-loads by GLOBAL image index; unwrap rank shards (Subset)
+    lookup = dataset
+    while isinstance(lookup, torch.utils.data.Subset):
+        lookup = lookup.dataset
 
     proto_epoch_dir = None
     if root_dir_for_saving_prototypes is not None:
@@ -159,7 +163,7 @@ def push_prototypes(dataloader,
 
     for s in range(0, len(my), bs):
         blk = my[s:s + bs]
-        imgs = [_item_image(dataset[img_idx]) for (_j, img_idx, _h, _w) in blk]
+        imgs = [_item_image(lookup[img_idx]) for (_j, img_idx, _h, _w) in blk]
         batch = torch.stack(imgs)
         if preprocess_input_function is not None:
             batch = preprocess_input_function(batch)
@@ -170,7 +174,7 @@ def push_prototypes(dataloader,
             updates[j] = feats[bi, :, h, w]
             have[j] = 1.0
             if proto_epoch_dir is not None:
-                _render_artifacts(dataset, img_idx, j, dist[bi, j],
+                _render_artifacts(lookup, img_idx, j, dist[bi, j],
                                   proto_epoch_dir,
                                   prototype_img_filename_prefix or 'prototype-img',
                                   preprocess_input_function)

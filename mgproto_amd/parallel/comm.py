@@ -73,9 +73,12 @@ class Comm:
         """All-gather equal-shaped tensors -> [world, *t.shape] (no host sync)."""
         if not self.is_distributed:
             return t.unsqueeze(0)
-        out = t.new_empty((self.world_size,) + tuple(t.shape))
-        dist.all_gather_into_tensor(out, t.contiguous())
-        return out
+        t = t.contiguous()
+        if t.dim() == 0:
+            t = t.unsqueeze(0)
+        out = t.new_empty((self.world_size * t.shape[0],) + tuple(t.shape[1:]))
+        dist.all_gather_into_tensor(out, t)
+        return out.view((self.world_size,) + tuple(t.shape))
 
     def all_gather_varlen(self, t: torch.Tensor) -> torch.Tensor:
         """All-gather 1-D tensors of differing lengths (host-syncing; used

@@ -1,0 +1,111 @@
+"""Data pipeline: PIL transforms, ImageFolder, preprocess."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+PIL = pytest.importorskip('PIL')
+from PIL import Image
+
+from mgproto_amd.data import transforms as T
+from mgproto_amd.data.folder import ImageFolder, MyImageFolder
+from mgproto_amd.data.preprocess import (preprocess_input_function,
+                                         undo_preprocess_input_function)
+
+
+def _img(w=64, h=48, seed=0):
+    rng = np.random.RandomState(seed)
+    return Image.fromarray(rng.randint(0, 255, (h, w, 3), dtype=np.uint8))
+
+
+def test_to_tensor_normalize_roundtrip():
+    img = _img()
+    t = T.ToTensor()(img)
+    assert t.shape == (3, 48, 64)
+    assert 0.0 <= t.min() and t.max() <= 1.0
+    x = t.unsqueeze(0)
+    y = preprocess_input_function(x)
+    back = undo_preprocess_input_function(y)
+    assert torch.allclose(back, x, atol=1e-6)
+
+
+def test_resize_semantics():
+    img = _img(100, 50)
+    r = T.Resize(25)(img)            # shorter side -> 25
+    assert r.size == (50, 25)
+    r2 = T.Resize((30, 40))(img)     # exact (h, w)
+    assert r2.size == (40, 30)
+
+
+def test_center_crop():
+    img = _img(100, 80)
+    c = T.CenterCrop(60)(img)
+    assert c.size == (60, 60)
+
+
+def test_random_transforms_run_and_preserve_size():
+    img = _img(64, 64)
+    tf = T.Compose([
+        T.RandomPerspective(distortion_scale=0.2, p=1.0),
+        T.ColorJitter((0.6, 1.4), (0.6, 1.4), (0.6, 1.4), (-0.02, 0.02)),
+        T.RandomHorizontalFlip(),
+        T.RandomAffine(degrees=25, shear=(-15, 15), translate=[0.05, 0.05]),
+        T.RandomResizedCrop(size=(32, 32), scale=(0.60, 1.0)),
+        T.ToTensor(),
+    ])
+    out = tf(img)
+    assert out.shape == (3, 32, 32)
+    assert torch.isfinite(out).all()
+
+
+def test_image_folder(tmp_path):
+    for ci, cls in enumerate(['001.sparrow', '002.wren']):
+        d = tmp_path / cls
+        d.mkdir()
+        for i in range(3):
+            _img(seed=ci * 10 + i).save(d / f'img_{i}.jpg')
+    ds = ImageFolder(str(tmp_path), T.Compose([T.ToTensor()]))
+    assert len(ds) == 6
+    assert ds.classes == ['001.sparrow', '002.wren']
+    img, label, idx = ds[4]
+    assert label == 1 and idx == 4
+    assert img.shape[0] == 3
+
+    mds = MyImageFolder(str(tmp_path), T.Compose([T.ToTensor()]))
+    (img, label), (path, plabel) = mds[0]
+    assert label == plabel == 0
+    assert path.endswith('.jpg')
+
+
+def test_push_accepts_reference_format(tmp_path):
+    """push_prototypes consumes the reference MyImageFolder item format."""
+    from torch.utils.data import DataLoader
+    from mgproto_amd.engine import push_prototypes
+    from mgproto_amd.model import construct_MGProto
+
+    for ci, cls in enumerate(['a', 'b']):
+        d = tmp_path / cls
+        d.mkdir()
+        for i in range(2):
+            _img(64, 64, seed=ci * 10 + i).save(d / f'{i}.jpg')
+    ds = MyImageFolder(str(tmp_path),
+                       T.Compose([T.Resize((64, 64)), T.ToTensor()]))
+
+    def collate(batch):
+        imgs = torch.stack([b[0][0] for b in batch])
+        labels = torch.tensor([b[0][1] for b in batch])
+        paths = [b[1][0] for b in batch]
+        return (imgs, labels), (paths, labels)
+
+    loader = DataLoader(ds, batch_size=2, collate_fn=collate)
+    torch.manual_seed(0)
+    model = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                              prototype_shape=(4, 16, 1, 1), num_classes=2,
+                              add_on_layers_type='regular', sz_embedding=8,
+                              mem_capacity=4, mine_K=2)
+    model.eval()
+    chosen = push_prototypes(loader, model, log=lambda *a: None,
+                             preprocess_input_function=preprocess_input_function)
+    assert len(chosen) > 0

@@ -1,0 +1,262 @@
+"""Multi-process DP correctness on CPU (gloo, world_size=2).
+
+Covers SURVEY.md §2.2 C1-C4: bucketed gradient all-reduce, DP-correct
+memory-bank enqueue (identical banks on all ranks), replicated EM
+determinism, distributed push candidate merge, and metric reduction —
+the paths the driver exercises on 8 GPUs with RCCL.
+"""
+
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _run_workers(fn, world=2, extra=()):
+    ctx = mp.get_context('spawn')
+    port = str(29600 + (os.getpid() + hash(fn.__name__)) % 200)
+    procs = []
+    q = ctx.SimpleQueue()
+    for rank in range(world):
+        p = ctx.Process(target=_worker_entry,
+                        args=(fn.__name__, rank, world, port, q, extra))
+        p.start()
+        procs.append(p)
+    results = {}
+    for _ in range(world):
+        rank, payload = q.get()
+        if isinstance(payload, str) and payload.startswith('ERROR'):
+            for p in procs:
+                p.terminate()
+            pytest.fail(f'rank {rank}: {payload}')
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=60)
+    return results
+
+
+def _worker_entry(fn_name, rank, world, port, q, extra):
+    os.environ['RANK'] = str(rank)
+    os.environ['WORLD_SIZE'] = str(world)
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = port
+    torch.manual_seed(100 + rank)
+    try:
+        payload = globals()[fn_name](rank, world, *extra)
+        q.put((rank, payload))
+    except Exception as e:  # noqa: BLE001
+        import traceback
+        q.put((rank, 'ERROR ' + traceback.format_exc()))
+        raise
+    finally:
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+# ---------------------------------------------------------------- reducer
+
+def test_grad_reducer_matches_mean():
+    _run_workers(impl_grad_reducer_matches_mean)
+
+
+def impl_grad_reducer_matches_mean(rank, world):
+    from mgproto_amd.parallel import Comm, BucketedGradReducer
+    comm = Comm(backend='gloo')
+    torch.manual_seed(0)  # same init everywhere
+    net = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.ReLU(),
+                              torch.nn.Linear(32, 4))
+    reducer = BucketedGradReducer(net, comm, bucket_mb=1)
+
+    torch.manual_seed(1000 + rank)  # different data per rank
+    x = torch.randn(8, 16)
+    y = torch.randn(8, 4)
+
+    reducer.prepare()
+    loss = ((net(x) - y) ** 2).mean()
+    loss.backward()
+    reducer.finalize()
+    got = [p.grad.clone() for p in net.parameters()]
+
+    # oracle: gather per-rank grads, average
+    net.zero_grad(set_to_none=True)
+    loss = ((net(x) - y) ** 2).mean()
+    loss.backward()
+    local = [p.grad.clone() for p in net.parameters()]
+    import torch.distributed as dist
+    want = []
+    for g in local:
+        t = g.clone()
+        dist.all_reduce(t)
+        want.append(t / world)
+    for g1, g2 in zip(got, want):
+        assert torch.allclose(g1, g2, atol=1e-6)
+    return 'ok'
+
+
+def test_grad_reducer_partial_bucket():
+    _run_workers(impl_grad_reducer_partial_bucket)
+
+
+def impl_grad_reducer_partial_bucket(rank, world):
+    """Frozen params mid-bucket must not hang or corrupt reduction."""
+    from mgproto_amd.parallel import Comm, BucketedGradReducer
+    comm = Comm(backend='gloo')
+    torch.manual_seed(0)
+    net = torch.nn.Sequential(torch.nn.Linear(8, 8), torch.nn.Linear(8, 8))
+    reducer = BucketedGradReducer(net, comm, bucket_mb=1024)  # one bucket
+    # freeze the second layer AFTER reducer construction (warm phase)
+    for p in net[1].parameters():
+        p.requires_grad = False
+    torch.manual_seed(2000 + rank)
+    x = torch.randn(4, 8)
+    reducer.prepare()
+    net(x).sum().backward()
+    reducer.finalize()
+    g = net[0].weight.grad.clone()
+    import torch.distributed as dist
+    want = net[0].weight.grad.clone()
+    dist.all_reduce(want)
+    want /= world
+    assert torch.allclose(g, want, atol=1e-6)
+    return 'ok'
+
+
+# ----------------------------------------------------------- enqueue + EM
+
+def _tiny_model():
+    from mgproto_amd.model import construct_MGProto
+    torch.manual_seed(0)
+    return construct_MGProto('resnet18', pretrained=False, img_size=64,
+                             prototype_shape=(20, 16, 1, 1), num_classes=5,
+                             add_on_layers_type='regular', sz_embedding=8,
+                             mem_capacity=6, mine_K=3)
+
+
+def test_distributed_enqueue_identical_banks():
+    _run_workers(impl_distributed_enqueue_identical_banks)
+
+
+def impl_distributed_enqueue_identical_banks(rank, world):
+    from mgproto_amd.parallel import Comm, make_dp_correct
+    comm = Comm(backend='gloo')
+    model = _tiny_model()
+    model = make_dp_correct(model, comm, train_batch_size=4)
+
+    torch.manual_seed(3000 + rank)
+    x = torch.randn(4, 3, 64, 64)
+    gt = torch.randint(0, 5, (4,))
+    with torch.no_grad():
+        model(x, gt)
+
+    # all ranks must hold bit-identical banks and dirty flags
+    import torch.distributed as dist
+    state = torch.cat([model.queue.mem.flatten(),
+                       model.queue.mem_len.float(),
+                       model.queue.head.float(),
+                       model.memory_updated_cls.float()])
+    ref = state.clone()
+    dist.broadcast(ref, src=0)
+    assert torch.equal(state, ref), 'bank state diverged across ranks'
+
+    # EM is replicated deterministic compute -> stays identical
+    model.queue.mem_len.fill_(model.queue.cap_cls)  # force full
+    model.update_GMM()
+    means = model.prototype_means.data.clone()
+    ref = means.clone()
+    dist.broadcast(ref, src=0)
+    assert torch.allclose(means, ref, atol=0), 'EM diverged across ranks'
+    return 'ok'
+
+
+def test_distributed_enqueue_matches_bigbatch():
+    _run_workers(impl_distributed_enqueue_matches_bigbatch)
+
+
+def impl_distributed_enqueue_matches_bigbatch(rank, world):
+    """2-rank enqueue of per-rank batches == single push of the rank-ordered
+    concatenation (the rank-invariance contract)."""
+    from mgproto_amd.parallel import Comm, DistributedEnqueue
+    from mgproto_amd.utils.memory import MemoryBank
+    comm = Comm(backend='gloo')
+    model = _tiny_model()
+    enq = DistributedEnqueue(model, comm, max_items=12)
+
+    torch.manual_seed(4000 + rank)
+    feats = torch.randn(7, 16)
+    labels = torch.randint(0, 5, (7,))
+    enq(feats, labels)
+
+    # oracle on every rank: gather both ranks' items, push in rank order
+    import torch.distributed as dist
+    all_f = [torch.zeros_like(feats) for _ in range(world)]
+    all_l = [torch.zeros_like(labels) for _ in range(world)]
+    dist.all_gather(all_f, feats)
+    dist.all_gather(all_l, labels)
+    oracle = MemoryBank(5, 16, capacity=5 * 6)
+    oracle.push(torch.cat(all_f), torch.cat(all_l))
+    for c in range(5):
+        assert torch.allclose(model.queue._logical(c), oracle._logical(c))
+    return 'ok'
+
+
+# ------------------------------------------------------------------- push
+
+def test_distributed_push_matches_single():
+    _run_workers(impl_distributed_push_matches_single)
+
+
+def impl_distributed_push_matches_single(rank, world):
+    from mgproto_amd.parallel import Comm
+    from mgproto_amd.engine import push_prototypes
+    from mgproto_amd.data import SyntheticImages
+    from torch.utils.data import DataLoader, Subset
+
+    comm = Comm(backend='gloo')
+    model = _tiny_model()
+    model.eval()
+    ds = SyntheticImages(n=8, num_classes=5, img_size=64, normalize=False)
+
+    def collate(batch):
+        return (torch.stack([b[0] for b in batch]),
+                torch.tensor([b[1] for b in batch]),
+                torch.tensor([b[2] for b in batch]))
+
+    # rank shard: interleaved indices (DistributedSampler-style)
+    shard = Subset(ds, list(range(rank, len(ds), world)))
+
+    def collate_shard(batch):
+        return collate(batch)
+
+    loader = DataLoader(shard, batch_size=4, collate_fn=collate_shard)
+    # Subset items keep their global idx in position 2 -> global identity ok
+    chosen = push_prototypes(loader, model, log=lambda *a: None, comm=comm)
+
+    # oracle: single-process push over the full dataset
+    model2 = _tiny_model()
+    model2.eval()
+    full_loader = DataLoader(ds, batch_size=4, collate_fn=collate)
+    chosen_single = push_prototypes(full_loader, model2, log=lambda *a: None)
+
+    assert chosen == chosen_single, (chosen, chosen_single)
+    assert torch.allclose(model.prototype_means.data,
+                          model2.prototype_means.data, atol=1e-5)
+    return 'ok'
+
+
+# ---------------------------------------------------------------- metrics
+
+def test_metric_allreduce():
+    _run_workers(impl_metric_allreduce)
+
+
+def impl_metric_allreduce(rank, world):
+    from mgproto_amd.parallel import Comm
+    comm = Comm(backend='gloo')
+    t = torch.tensor([float(rank + 1), 10.0])
+    out = comm.all_reduce_sum(t.clone())
+    assert torch.allclose(out, torch.tensor([3.0, 20.0]))
+    v = comm.all_gather_varlen(torch.arange(rank + 1).float())
+    assert v.tolist() == [0.0, 0.0, 1.0]
+    return 'ok'

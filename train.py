@@ -1,0 +1,246 @@
+#!/usr/bin/env python3
+"""Training driver — the reference ``main.py`` rebuilt for MI355X.
+
+Single GPU:  python train.py -arch resnet34 -dataset CUB
+8-GPU node:  python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+                 --master-addr 127.0.0.1 train.py -arch resnet50 ...
+
+Differences from the reference driver (main.py):
+* one process per GPU over RCCL instead of DataParallel (main.py:184);
+* real input pipeline (num_workers > 0; the reference ran the loader on the
+  main thread, main.py:94);
+* full checkpoint/resume (the reference can only save);
+* works on synthetic data out of the box (-dataset synthetic) since this
+  environment has no datasets.
+"""
+
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+from torch.utils.data import DataLoader
+
+from mgproto_amd import settings
+from mgproto_amd.settings import Settings
+from mgproto_amd.model import construct_MGProto
+from mgproto_amd.losses import build_aux_loss
+from mgproto_amd.engine import trainer as tnt
+from mgproto_amd.engine.push import push_prototypes
+from mgproto_amd.engine.trainer import EMRunner
+from mgproto_amd.parallel import Comm, BucketedGradReducer, make_dp_correct
+from mgproto_amd.utils import makedir, datestr, create_logger, MetricsLogger
+from mgproto_amd.utils.checkpoint import (save_model_w_condition,
+                                          save_train_state, load_train_state)
+from mgproto_amd.data.synthetic import SyntheticImages
+
+
+def build_loaders(cfg: Settings, comm):
+    """Train/push/test/ood loaders. Real image folders when the dirs exist,
+    synthetic otherwise."""
+    from mgproto_amd.data import loaders as L
+    world = comm.world_size if comm else 1
+    rank = comm.rank if comm else 0
+    if os.path.isdir(cfg.train_dir):
+        return L.build_image_loaders(cfg, world=world, rank=rank)
+    # synthetic fallback (no datasets in this environment)
+    n = 1600
+    train_ds = SyntheticImages(n, cfg.num_classes, cfg.img_size)
+    push_ds = SyntheticImages(n, cfg.num_classes, cfg.img_size, normalize=False)
+    test_ds = SyntheticImages(n // 4, cfg.num_classes, cfg.img_size, seed=1)
+    ood1 = SyntheticImages(n // 8, cfg.num_classes, cfg.img_size, seed=2)
+    ood2 = SyntheticImages(n // 8, cfg.num_classes, cfg.img_size, seed=3)
+
+    def shard(ds):
+        if world > 1:
+            from torch.utils.data import Subset
+            return Subset(ds, list(range(rank, len(ds), world)))
+        return ds
+
+    def collate(batch):
+        return (torch.stack([b[0] for b in batch]),
+                torch.tensor([b[1] for b in batch]),
+                torch.tensor([b[2] for b in batch]))
+
+    mk = lambda ds, bs: DataLoader(ds, batch_size=bs, num_workers=cfg.num_workers,  # noqa: E731
+                                   collate_fn=collate, drop_last=False)
+    return (mk(shard(train_ds), cfg.train_batch_size),
+            mk(shard(push_ds), cfg.train_push_batch_size),
+            mk(shard(test_ds), cfg.test_batch_size),
+            mk(shard(ood1), cfg.test_batch_size),
+            mk(shard(ood2), cfg.test_batch_size))
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument('-gpuid', type=str, default='0')      # reference compat
+    parser.add_argument('-dataset', type=str, default='CUB')
+    parser.add_argument('-arch', type=str, default='resnet34')
+    parser.add_argument('-aux_loss', type=str, default='Proxy_Anchor')
+    parser.add_argument('-aux_emb_sz', type=int, default=32)
+    parser.add_argument('-mem_sz', type=int, default=800)
+    parser.add_argument('-mine_level', type=int, default=20)
+    parser.add_argument('--epochs', type=int, default=None)
+    parser.add_argument('--resume', type=str, default=None)
+    parser.add_argument('--out', type=str, default=None)
+    parser.add_argument('--ood-eval', action='store_true')
+    parser.add_argument('--addon', type=str, default=None)
+    args = parser.parse_args()
+
+    cfg = Settings(base_architecture=args.arch, aux_loss=args.aux_loss,
+                   sz_embedding=args.aux_emb_sz, mem_capacity=args.mem_sz,
+                   mine_K=args.mine_level)
+    if args.addon:
+        cfg.add_on_layers_type = args.addon
+    if args.epochs is not None:
+        cfg.num_train_epochs = args.epochs
+        cfg.push_epochs = [i for i in range(args.epochs) if i % 10 == 0]
+
+    comm = Comm() if int(os.environ.get('WORLD_SIZE', '1')) > 1 else None
+    rank = comm.rank if comm else 0
+    device = (comm.device if comm
+              else (torch.device('cuda', 0) if torch.cuda.is_available()
+                    else torch.device('cpu')))
+    if device.type == 'cuda':
+        torch.cuda.set_device(device)
+        torch.backends.cudnn.benchmark = True
+
+    model_dir = args.out or 'saved_models/{}/{}/'.format(args.arch, datestr())
+    if rank == 0:
+        makedir(model_dir)
+        makedir(os.path.join(model_dir, 'img'))
+    if comm:
+        comm.barrier()
+    log, logclose = (create_logger(os.path.join(model_dir, 'train.log'))
+                     if rank == 0 else (lambda s: None, lambda: None))
+    metrics = MetricsLogger(os.path.join(model_dir, 'metrics.jsonl'), rank=rank)
+
+    loaders = build_loaders(cfg, comm)
+    train_loader, push_loader, test_loader, ood1_loader, ood2_loader = loaders
+
+    ppnet = construct_MGProto(
+        base_architecture=args.arch, pretrained=True, img_size=cfg.img_size,
+        prototype_shape=cfg.prototype_shape, num_classes=cfg.num_classes,
+        prototype_activation_function=cfg.prototype_activation_function,
+        add_on_layers_type=cfg.add_on_layers_type,
+        sz_embedding=args.aux_emb_sz, mem_capacity=args.mem_sz,
+        mine_K=args.mine_level).to(device)
+    if device.type == 'cuda' and cfg.channels_last:
+        ppnet.features = ppnet.features.to(memory_format=torch.channels_last)
+
+    aux_criterion = build_aux_loss(args.aux_loss, nb_classes=cfg.num_classes,
+                                   sz_embed=args.aux_emb_sz, mrg=0.1,
+                                   beta=32).to(device)
+
+    joint_optimizer = torch.optim.Adam([
+        {'params': ppnet.features.parameters(),
+         'lr': cfg.joint_optimizer_lrs['features'], 'weight_decay': 1e-4},
+        {'params': ppnet.add_on_layers.parameters(),
+         'lr': cfg.joint_optimizer_lrs['add_on_layers'], 'weight_decay': 1e-4},
+        {'params': aux_criterion.parameters(),
+         'lr': cfg.joint_optimizer_lrs['features'] * 100, 'weight_decay': 1e-4},
+    ])
+    joint_lr_scheduler = torch.optim.lr_scheduler.StepLR(
+        joint_optimizer, step_size=1, gamma=cfg.joint_lr_gamma)
+    warm_optimizer = torch.optim.Adam([
+        {'params': ppnet.add_on_layers.parameters(),
+         'lr': cfg.warm_optimizer_lrs['add_on_layers'], 'weight_decay': 1e-4},
+        {'params': aux_criterion.parameters(),
+         'lr': cfg.joint_optimizer_lrs['features'] * 100, 'weight_decay': 1e-4},
+    ])
+    ppnet.prototype_lr = cfg.joint_optimizer_lrs['prototype_vectors']
+
+    reducer = None
+    if comm is not None and comm.is_distributed:
+        ppnet = make_dp_correct(ppnet, comm, cfg.train_batch_size)
+        comm.broadcast_module(aux_criterion)
+        reducer = BucketedGradReducer([ppnet, aux_criterion], comm,
+                                      bucket_mb=cfg.grad_bucket_mb)
+
+    em_runner = EMRunner(ppnet, use_stream=(device.type == 'cuda'
+                                            and cfg.em_stream))
+    start_epoch = 0
+    if args.resume:
+        state = load_train_state(args.resume, ppnet,
+                                 {'joint': joint_optimizer, 'warm': warm_optimizer},
+                                 {'joint_lr': joint_lr_scheduler},
+                                 map_location=device)
+        start_epoch = state['epoch'] + 1
+        log(f'resumed from {args.resume} at epoch {start_epoch}')
+
+    log('start training')
+    decay_epochs = cfg.lr_decay_epochs()
+    epoch = start_epoch
+    for epoch in range(start_epoch, cfg.num_train_epochs):
+        log('epoch: \t{0}'.format(epoch))
+        use_mining = epoch >= cfg.mine_start
+        update_GMM = (epoch >= cfg.updateGMM_start
+                      and bool((ppnet.queue.mem_len == ppnet.capacity_pc).all()))
+        log('use mining: \t{0}'.format(use_mining))
+        log('update GMM: \t{0}'.format(update_GMM))
+
+        kw = dict(device=device, amp_dtype=cfg.amp_dtype, comm=comm,
+                  metrics=metrics)
+        if epoch < cfg.num_warm_epochs:
+            tnt.warm_only(ppnet, log=log)
+            _, train_results = tnt.train(
+                ppnet, train_loader, warm_optimizer, aux_criterion=aux_criterion,
+                use_mine=use_mining, update_GMM=update_GMM, coefs=cfg.coefs,
+                log=log, em_runner=em_runner, **kw)
+        else:
+            tnt.joint(ppnet, log=log)
+            if epoch in decay_epochs:
+                joint_lr_scheduler.step()
+            _, train_results = tnt.train(
+                ppnet, train_loader, joint_optimizer, aux_criterion=aux_criterion,
+                use_mine=use_mining, update_GMM=update_GMM, coefs=cfg.coefs,
+                log=log, em_runner=em_runner, **kw)
+
+        if args.ood_eval:
+            accu, _ = tnt.test(ppnet, (test_loader, ood1_loader, ood2_loader),
+                               log=log, ood=True, **kw)
+        else:
+            accu, _ = tnt.test(ppnet, (test_loader,), log=log, **kw)
+        if rank == 0:
+            save_model_w_condition(ppnet, model_dir, str(epoch) + 'nopush',
+                                   accu, 0.00, log=log)
+            save_train_state(os.path.join(model_dir, 'latest.pth'), ppnet,
+                             {'joint': joint_optimizer, 'warm': warm_optimizer},
+                             {'joint_lr': joint_lr_scheduler}, epoch)
+        metrics.log({'epoch': epoch,
+                     'lr': joint_optimizer.param_groups[0]['lr'],
+                     'update_GMM': update_GMM * 1.0,
+                     'use_mining': use_mining * 1.0})
+
+        if epoch >= cfg.push_start and epoch in cfg.push_epochs:
+            push_prototypes(
+                push_loader, ppnet, class_specific=True,
+                preprocess_input_function=_preprocess(),
+                root_dir_for_saving_prototypes=os.path.join(model_dir, 'img'),
+                epoch_number=epoch, prototype_img_filename_prefix='prototype-img',
+                log=log, comm=comm, device=device)
+            accu, _ = tnt.test(ppnet, (test_loader,), log=log, **kw)
+            if rank == 0:
+                save_model_w_condition(ppnet, model_dir, str(epoch) + 'push',
+                                       accu, 0.00, log=log)
+
+    # final pruning (reference main.py:285)
+    ppnet.prune_prototypes_topM(top_M=8)
+    accu, _ = tnt.test(ppnet, (test_loader,), log=log, device=device,
+                       amp_dtype=cfg.amp_dtype, comm=comm)
+    if rank == 0:
+        save_model_w_condition(ppnet, model_dir, str(epoch) + 'prune', accu,
+                               0.00, log=log)
+    metrics.close()
+    logclose()
+
+
+def _preprocess():
+    from mgproto_amd.data.preprocess import preprocess_input_function
+    return preprocess_input_function
+
+
+if __name__ == '__main__':
+    main()
