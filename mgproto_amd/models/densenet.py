@@ -21,6 +21,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .resnet import PRETRAINED_DIR
+from .fused_bn import FusedBatchNorm2d
 
 
 class _DenseLayer(nn.Sequential):
@@ -28,12 +29,14 @@ class _DenseLayer(nn.Sequential):
 
     def __init__(self, num_input_features, growth_rate, bn_size, drop_rate):
         super().__init__()
-        self.add_module('norm1', nn.BatchNorm2d(num_input_features))
-        self.add_module('relu1', nn.ReLU(inplace=True))
+        self.add_module('norm1', FusedBatchNorm2d(num_input_features,
+                                                  fused_relu=True))
+        self.add_module('relu1', nn.Identity())
         self.add_module('conv1', nn.Conv2d(num_input_features, bn_size * growth_rate,
                                            kernel_size=1, stride=1, bias=False))
-        self.add_module('norm2', nn.BatchNorm2d(bn_size * growth_rate))
-        self.add_module('relu2', nn.ReLU(inplace=True))
+        self.add_module('norm2', FusedBatchNorm2d(bn_size * growth_rate,
+                                                  fused_relu=True))
+        self.add_module('relu2', nn.Identity())
         self.add_module('conv2', nn.Conv2d(bn_size * growth_rate, growth_rate,
                                            kernel_size=3, stride=1, padding=1, bias=False))
         self.drop_rate = drop_rate
@@ -73,8 +76,9 @@ class _Transition(nn.Sequential):
 
     def __init__(self, num_input_features, num_output_features):
         super().__init__()
-        self.add_module('norm', nn.BatchNorm2d(num_input_features))
-        self.add_module('relu', nn.ReLU(inplace=True))
+        self.add_module('norm', FusedBatchNorm2d(num_input_features,
+                                                 fused_relu=True))
+        self.add_module('relu', nn.Identity())
         self.add_module('conv', nn.Conv2d(num_input_features, num_output_features,
                                           kernel_size=1, stride=1, bias=False))
         self.add_module('pool', nn.AvgPool2d(kernel_size=2, stride=2))
@@ -97,8 +101,8 @@ class DenseNetFeatures(nn.Module):
         self.features = nn.Sequential(OrderedDict([
             ('conv0', nn.Conv2d(3, num_init_features, kernel_size=7, stride=2,
                                 padding=3, bias=False)),
-            ('norm0', nn.BatchNorm2d(num_init_features)),
-            ('relu0', nn.ReLU(inplace=True)),
+            ('norm0', FusedBatchNorm2d(num_init_features, fused_relu=True)),
+            ('relu0', nn.Identity()),
             # no pool0: stride stays 16 overall (reference densenet_features.py:116)
         ]))
         self.kernel_sizes.append(7)
@@ -127,8 +131,9 @@ class DenseNetFeatures(nn.Module):
                 self.features.add_module('transition%d' % (i + 1), trans)
                 num_features = num_features // 2
 
-        self.features.add_module('norm5', nn.BatchNorm2d(num_features))
-        self.features.add_module('final_relu', nn.ReLU(inplace=True))
+        self.features.add_module('norm5', FusedBatchNorm2d(num_features,
+                                                           fused_relu=True))
+        self.features.add_module('final_relu', nn.Identity())
         self.out_channels = num_features
 
         for m in self.modules():

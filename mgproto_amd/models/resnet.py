@@ -28,6 +28,8 @@ import warnings
 import torch
 import torch.nn as nn
 
+from .fused_bn import bn_act
+
 PRETRAINED_DIR = os.environ.get('MGPROTO_PRETRAINED_DIR', './pretrained_models')
 
 
@@ -55,13 +57,15 @@ class BasicBlock(nn.Module):
         self.stride = stride
 
     def forward(self, x):
-        identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
+        # bn+relu and bn+add+relu run as single fused NHWC kernels on GPU
+        # (models/fused_bn.py); the fallback path is numerically identical
+        out = bn_act(self.conv1(x), self.bn1, relu=True)
+        out = self.conv2(out)
         if self.downsample is not None:
-            identity = self.downsample(x)
-        out = self.relu(out + identity)
-        return out
+            identity = bn_act(self.downsample[0](x), self.downsample[1])
+        else:
+            identity = x
+        return bn_act(out, self.bn2, relu=True, residual=identity)
 
     def block_conv_info(self):
         return [3, 3], [self.stride, 1], [1, 1]
@@ -84,14 +88,14 @@ class Bottleneck(nn.Module):
         self.stride = stride
 
     def forward(self, x):
-        identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
+        out = bn_act(self.conv1(x), self.bn1, relu=True)
+        out = bn_act(self.conv2(out), self.bn2, relu=True)
+        out = self.conv3(out)
         if self.downsample is not None:
-            identity = self.downsample(x)
-        out = self.relu(out + identity)
-        return out
+            identity = bn_act(self.downsample[0](x), self.downsample[1])
+        else:
+            identity = x
+        return bn_act(out, self.bn3, relu=True, residual=identity)
 
     def block_conv_info(self):
         return [1, 3, 1], [1, self.stride, 1], [0, 1, 0]
@@ -156,7 +160,7 @@ class ResNetFeatures(nn.Module):
         return nn.Sequential(*blocks)
 
     def forward(self, x):
-        x = self.relu(self.bn1(self.conv1(x)))
+        x = bn_act(self.conv1(x), self.bn1, relu=True)
         # NOTE: no max-pool here — output stride is 16 (reference behaviour)
         x = self.layer1(x)
         x = self.layer2(x)

@@ -15,6 +15,7 @@ import torch
 import torch.nn as nn
 
 from .resnet import PRETRAINED_DIR
+from .fused_bn import FusedBatchNorm2d
 
 cfg = {
     'A': [64, 'M', 128, 'M', 256, 256, 'M', 512, 512, 'M', 512, 512, 'M'],
@@ -55,7 +56,8 @@ class VGGFeatures(nn.Module):
             else:
                 conv2d = nn.Conv2d(in_channels, v, kernel_size=3, padding=1)
                 if batch_norm:
-                    layers += [conv2d, nn.BatchNorm2d(v), nn.ReLU(inplace=True)]
+                    layers += [conv2d, FusedBatchNorm2d(v, fused_relu=True),
+                               nn.Identity()]
                 elif i >= len(layer_cfg) - 2 and not final_relu:
                     layers += [conv2d]
                 else:

@@ -10,7 +10,8 @@ import shutil
 import sys
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
-SRC = [os.path.join(_HERE, 'hip', 'mgproto_kernels.hip')]
+SRC = [os.path.join(_HERE, 'hip', 'mgproto_kernels.hip'),
+       os.path.join(_HERE, 'hip', 'fused_bn.hip')]
 OUT = os.path.join(_HERE, '_mgproto_hip.so')
 
 

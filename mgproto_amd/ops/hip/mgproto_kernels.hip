@@ -534,9 +534,25 @@ std::vector<torch::Tensor> argmax_hw(torch::Tensor probs) {
     return {vals, idx};
 }
 
+// fused BatchNorm kernels (fused_bn.hip)
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
+                                  torch::Tensor bias,
+                                  torch::Tensor running_mean,
+                                  torch::Tensor running_var,
+                                  bool training, double momentum, double eps,
+                                  bool relu,
+                                  c10::optional<torch::Tensor> residual);
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
+                                  torch::Tensor x, torch::Tensor weight,
+                                  torch::Tensor save_mean,
+                                  torch::Tensor save_rstd,
+                                  bool training, bool relu, bool has_res);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gmm_fwd", &gmm_fwd, "fused GMM score GEMM forward (MFMA f32)");
     m.def("gmm_bwd", &gmm_bwd, "GMM score GEMM backward wrt features");
     m.def("topk_hw", &topk_hw, "per-(b,p) top-T over HW with indices");
     m.def("argmax_hw", &argmax_hw, "per-(b,p) argmax over HW");
+    m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+Add)(+ReLU) forward, NHWC bf16");
+    m.def("bn_bwd", &bn_bwd, "fused BatchNorm(+Add)(+ReLU) backward, NHWC bf16");
 }

@@ -1,0 +1,155 @@
+"""Fused BatchNorm(+Add)(+ReLU) parity vs the plain-PyTorch path (GPU)."""
+
+import os
+
+import pytest
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+
+def _pair(N=4, C=64, H=14, W=14, seed=0, res=False):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(N, C, H, W, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    r = None
+    if res:
+        r = torch.randn(N, C, H, W, generator=g).to('cuda', torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+    return x, r
+
+
+def _ref_bn(x, bn, relu, residual, training):
+    y = F.batch_norm(x.float(), bn.running_mean.clone(), bn.running_var.clone(),
+                     bn.weight.float(), bn.bias.float(), training,
+                     bn.momentum, bn.eps)
+    if residual is not None:
+        y = y + residual.float()
+    return F.relu(y) if relu else y
+
+
+@pytest.mark.parametrize('relu,res,training',
+                         [(False, False, True), (True, False, True),
+                          (True, True, True), (False, True, True),
+                          (True, False, False), (False, False, False)])
+def test_bn_forward_parity(relu, res, training):
+    from mgproto_amd.models.fused_bn import bn_act
+    torch.manual_seed(0)
+    C = 64
+    bn = nn.BatchNorm2d(C).cuda()
+    bn.weight.data.uniform_(0.5, 1.5)
+    bn.bias.data.uniform_(-0.5, 0.5)
+    bn.running_mean.data.uniform_(-0.2, 0.2)
+    bn.running_var.data.uniform_(0.8, 1.2)
+    bn.train(training)
+
+    x, r = _pair(res=res)
+    rm0, rv0 = bn.running_mean.clone(), bn.running_var.clone()
+    want = _ref_bn(x, bn, relu, r, training)
+
+    with torch.no_grad():
+        got = bn_act(x, bn, relu=relu, residual=r)
+    assert got.dtype == torch.bfloat16
+    assert torch.allclose(got.float(), want, atol=3e-2, rtol=3e-2), \
+        (got.float() - want).abs().max().item()
+
+    if training:
+        # running stats updated like torch BN
+        M = x.numel() // C
+        xf = x.float().permute(0, 2, 3, 1).reshape(-1, C)
+        mean = xf.mean(0)
+        var = xf.var(0, unbiased=True)
+        want_rm = (1 - bn.momentum) * rm0 + bn.momentum * mean
+        want_rv = (1 - bn.momentum) * rv0 + bn.momentum * var
+        assert torch.allclose(bn.running_mean, want_rm, atol=1e-2, rtol=1e-2)
+        assert torch.allclose(bn.running_var, want_rv, atol=1e-2, rtol=1e-2)
+
+
+@pytest.mark.parametrize('relu,res', [(False, False), (True, False),
+                                      (True, True)])
+def test_bn_backward_parity(relu, res):
+    from mgproto_amd.models.fused_bn import bn_act
+    torch.manual_seed(1)
+    C = 64
+    bn1 = nn.BatchNorm2d(C).cuda().train()
+    bn1.weight.data.uniform_(0.5, 1.5)
+    bn1.bias.data.uniform_(-0.5, 0.5)
+    bn2 = nn.BatchNorm2d(C).cuda().train()
+    bn2.load_state_dict(bn1.state_dict())
+
+    x, r = _pair(seed=2, res=res)
+    x1 = x.clone().requires_grad_(True)
+    r1 = r.clone().requires_grad_(True) if res else None
+    y1 = bn_act(x1, bn1, relu=relu, residual=r1)
+    g = torch.randn_like(y1)
+    y1.backward(g)
+
+    # reference: fp32 autograd through unfused ops
+    x2 = x.float().clone().requires_grad_(True)
+    r2 = r.float().clone().requires_grad_(True) if res else None
+    y2 = F.batch_norm(x2, bn2.running_mean, bn2.running_var,
+                      bn2.weight, bn2.bias, True, bn2.momentum, bn2.eps)
+    if res:
+        y2 = y2 + r2
+    if relu:
+        y2 = F.relu(y2)
+    y2.backward(g.float())
+
+    assert torch.allclose(y1.float(), y2, atol=3e-2, rtol=3e-2)
+    assert torch.allclose(x1.grad.float(), x2.grad, atol=3e-2, rtol=3e-2), \
+        (x1.grad.float() - x2.grad).abs().max().item()
+    assert torch.allclose(bn1.weight.grad, bn2.weight.grad, atol=2e-1,
+                          rtol=2e-2)
+    assert torch.allclose(bn1.bias.grad, bn2.bias.grad, atol=2e-1, rtol=2e-2)
+    if res:
+        assert torch.allclose(r1.grad.float(), r2.grad, atol=3e-2, rtol=3e-2)
+
+
+def test_backbone_fused_vs_unfused():
+    """ResNet-18 trunk: fused path vs MGPROTO_NO_FUSED_BN fallback."""
+    from mgproto_amd.models import resnet18_features
+    torch.manual_seed(0)
+    net = resnet18_features().cuda().to(memory_format=torch.channels_last)
+    net.train()
+    x = torch.randn(2, 3, 64, 64).cuda().contiguous(
+        memory_format=torch.channels_last)
+
+    os.environ['MGPROTO_NO_FUSED_BN'] = '1'
+    try:
+        with torch.autocast('cuda', dtype=torch.bfloat16), torch.no_grad():
+            want = net(x).float()
+    finally:
+        os.environ.pop('MGPROTO_NO_FUSED_BN')
+    with torch.autocast('cuda', dtype=torch.bfloat16), torch.no_grad():
+        got = net(x).float()
+    assert torch.allclose(got, want, atol=1e-1, rtol=5e-2), \
+        (got - want).abs().max().item()
+
+
+def test_fused_bn_graph_capturable():
+    from mgproto_amd.models.fused_bn import bn_act
+    torch.manual_seed(0)
+    bn = nn.BatchNorm2d(64).cuda().train()
+    x, r = _pair(res=True)
+    x = x.clone().requires_grad_(True)
+
+    def fn():
+        y = bn_act(x, bn, relu=True, residual=r)
+        y.float().sum().backward()
+        x.grad = None
+        bn.weight.grad = None
+        bn.bias.grad = None
+
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):
+            fn()
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        fn()
+    g.replay()
+    torch.cuda.synchronize()
