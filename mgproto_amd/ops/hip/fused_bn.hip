@@ -44,27 +44,61 @@ __device__ __forceinline__ short f2bf(float f) {
 // forward: per-channel sums (training stats)
 // ---------------------------------------------------------------------------
 
+// Each block covers rows_per_iter = 256/(C/8) rows per iteration (all 256
+// threads active for any C), accumulates per-thread partials in registers,
+// combines row-groups through LDS atomics, and emits ONE global atomic per
+// channel per block.  Wave lanes read consecutive rows' consecutive 16-byte
+// chunks -> fully coalesced.  For C/8 >= 256 the channel axis alone fills
+// the block (grid.y splits channels).
 __global__ __launch_bounds__(256)
 void bn_stats_kernel(const short* __restrict__ x, float* __restrict__ sums,
                      long M, int C) {
-    const int c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
-    if (c8 >= C) return;
-    float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (long m = blockIdx.x; m < M; m += gridDim.x) {
-        const short8 v = *reinterpret_cast<const short8*>(x + m * C + c8);
+    extern __shared__ __attribute__((aligned(16))) float lds[];  // [2C]
+    const int tpr = C / 8;
+    if (tpr >= 256) {                       // channel-split path, no sharing
+        const int c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
+        if (c8 >= C) return;
+        float s[8] = {0}, q[8] = {0};
+        for (long m = blockIdx.x; m < M; m += gridDim.x) {
+            const short8 v = *reinterpret_cast<const short8*>(x + m * C + c8);
+            #pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                const float f = bf2f(v[i]);
+                s[i] += f; q[i] += f * f;
+            }
+        }
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
-            const float f = bf2f(v[i]);
-            s[i] += f;
-            q[i] += f * f;
+            atomicAdd(&sums[c8 + i], s[i]);
+            atomicAdd(&sums[C + c8 + i], q[i]);
+        }
+        return;
+    }
+    for (int i = threadIdx.x; i < 2 * C; i += 256) lds[i] = 0.f;
+    __syncthreads();
+    const int rpi = 256 / tpr;
+    const int rsub = threadIdx.x / tpr;
+    const int c8 = (threadIdx.x % tpr) * 8;
+    if (rsub < rpi) {
+        float s[8] = {0}, q[8] = {0};
+        for (long m = (long)blockIdx.x * rpi + rsub; m < M;
+             m += (long)gridDim.x * rpi) {
+            const short8 v = *reinterpret_cast<const short8*>(x + m * C + c8);
+            #pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                const float f = bf2f(v[i]);
+                s[i] += f; q[i] += f * f;
+            }
+        }
+        #pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            atomicAdd(&lds[c8 + i], s[i]);
+            atomicAdd(&lds[C + c8 + i], q[i]);
         }
     }
-    #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-        atomicAdd(&sums[c8 + i], s[i]);
-        atomicAdd(&sums[C + c8 + i], q[i]);
-    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < 2 * C; i += 256)
+        atomicAdd(&sums[i], lds[i]);
 }
 
 // finalize: mean/rstd, running-stat update, scale/shift for the apply pass
@@ -161,35 +195,74 @@ void bn_bwd_reduce_kernel(const short* __restrict__ dy,
                           const float* __restrict__ save_rstd,
                           float* __restrict__ sums,  // [2C]
                           long M, int C) {
-    const int c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
-    if (c8 >= C) return;
-    float sd[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    float sx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    float mean[8], rstd[8];
-    #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-        mean[i] = save_mean[c8 + i];
-        rstd[i] = save_rstd[c8 + i];
-    }
-    for (long m = blockIdx.x; m < M; m += gridDim.x) {
-        const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
-        const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
-        short8 yv;
-        if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+    extern __shared__ __attribute__((aligned(16))) float lds[];  // [2C]
+    const int tpr = C / 8;
+    if (tpr >= 256) {
+        const int c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
+        if (c8 >= C) return;
+        float sd[8] = {0}, sx[8] = {0};
+        float mean[8], rstd[8];
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
-            float gf = bf2f(g[i]);
-            if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
-            const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
-            sd[i] += gf;
-            sx[i] += gf * xhat;
+            mean[i] = save_mean[c8 + i];
+            rstd[i] = save_rstd[c8 + i];
+        }
+        for (long m = blockIdx.x; m < M; m += gridDim.x) {
+            const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
+            const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
+            short8 yv;
+            if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+            #pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                float gf = bf2f(g[i]);
+                if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
+                const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
+                sd[i] += gf; sx[i] += gf * xhat;
+            }
+        }
+        #pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            atomicAdd(&sums[c8 + i], sd[i]);
+            atomicAdd(&sums[C + c8 + i], sx[i]);
+        }
+        return;
+    }
+    for (int i = threadIdx.x; i < 2 * C; i += 256) lds[i] = 0.f;
+    __syncthreads();
+    const int rpi = 256 / tpr;
+    const int rsub = threadIdx.x / tpr;
+    const int c8 = (threadIdx.x % tpr) * 8;
+    if (rsub < rpi) {
+        float sd[8] = {0}, sx[8] = {0};
+        float mean[8], rstd[8];
+        #pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            mean[i] = save_mean[c8 + i];
+            rstd[i] = save_rstd[c8 + i];
+        }
+        for (long m = (long)blockIdx.x * rpi + rsub; m < M;
+             m += (long)gridDim.x * rpi) {
+            const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
+            const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
+            short8 yv;
+            if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+            #pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                float gf = bf2f(g[i]);
+                if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
+                const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
+                sd[i] += gf; sx[i] += gf * xhat;
+            }
+        }
+        #pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            atomicAdd(&lds[c8 + i], sd[i]);
+            atomicAdd(&lds[C + c8 + i], sx[i]);
         }
     }
-    #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-        atomicAdd(&sums[c8 + i], sd[i]);
-        atomicAdd(&sums[C + c8 + i], sx[i]);
-    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < 2 * C; i += 256)
+        atomicAdd(&sums[i], lds[i]);
 }
 
 // apply: dx = gamma*rstd * (dy_eff - sum_dy/M - xhat*sum_dyxhat/M)
@@ -290,12 +363,19 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
     auto scale_shift = torch::empty({2 * C}, fopt);
     auto y = torch::empty_like(x);
 
-    const int grid_y = cdiv(C / 8, 256);
+    const int tpr = C / 8;
+    const int grid_y = tpr >= 256 ? cdiv(tpr, 256) : 1;
+    const int lds_bytes = 2 * C * sizeof(float);
+    auto reduce_grid = [&]() {
+        if (tpr >= 256)
+            return dim3(std::max(1, std::min((int)M, 4096 / grid_y)), grid_y);
+        const int rpi = 256 / tpr;
+        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 1280)), 1);
+    };
     if (training) {
         auto sums = torch::zeros({2 * C}, fopt);
-        const int grid_x = std::max(1, std::min((int)M, 4096 / grid_y));
-        hipLaunchKernelGGL(bn_stats_kernel, dim3(grid_x, grid_y), dim3(256),
-                           0, stream, (const short*)x.data_ptr(),
+        hipLaunchKernelGGL(bn_stats_kernel, reduce_grid(), dim3(256),
+                           lds_bytes, stream, (const short*)x.data_ptr(),
                            sums.data_ptr<float>(), M, C);
         hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256),
                            0, stream, sums.data_ptr<float>(),
@@ -351,15 +431,22 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
     auto dres = has_res ? torch::empty_like(x) : torch::empty({0}, x.options());
     auto sums = torch::zeros({2 * C}, fopt);
 
-    const int grid_y = cdiv(C / 8, 256);
+    const int tpr = C / 8;
+    const int grid_y = tpr >= 256 ? cdiv(tpr, 256) : 1;
+    const int lds_bytes = 2 * C * sizeof(float);
+    auto reduce_grid = [&]() {
+        if (tpr >= 256)
+            return dim3(std::max(1, std::min((int)M, 4096 / grid_y)), grid_y);
+        const int rpi = 256 / tpr;
+        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 1280)), 1);
+    };
     const long total = M * (C / 8);
     const int blocks = (int)std::min<long>(cdiv(total, 256), 2048);
 
     if (training) {
-        const int grid_x = std::max(1, std::min((int)M, 4096 / grid_y));
         if (relu)
             hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>),
-                               dim3(grid_x, grid_y), dim3(256), 0, stream,
+                               reduce_grid(), dim3(256), lds_bytes, stream,
                                (const short*)dy.data_ptr(),
                                (const short*)y.data_ptr(),
                                (const short*)x.data_ptr(),
@@ -368,7 +455,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                sums.data_ptr<float>(), M, C);
         else
             hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>),
-                               dim3(grid_x, grid_y), dim3(256), 0, stream,
+                               reduce_grid(), dim3(256), lds_bytes, stream,
                                (const short*)dy.data_ptr(),
                                (const short*)y.data_ptr(),
                                (const short*)x.data_ptr(),
@@ -397,8 +484,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
         // reduce (xhat uses running stats)
         if (relu)
             hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>),
-                               dim3(std::max(1, std::min((int)M, 4096 / grid_y)),
-                                    grid_y), dim3(256), 0, stream,
+                               reduce_grid(), dim3(256), lds_bytes, stream,
                                (const short*)dy.data_ptr(),
                                (const short*)y.data_ptr(),
                                (const short*)x.data_ptr(),
@@ -407,8 +493,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                sums.data_ptr<float>(), M, C);
         else
             hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>),
-                               dim3(std::max(1, std::min((int)M, 4096 / grid_y)),
-                                    grid_y), dim3(256), 0, stream,
+                               reduce_grid(), dim3(256), lds_bytes, stream,
                                (const short*)dy.data_ptr(),
                                (const short*)y.data_ptr(),
                                (const short*)x.data_ptr(),
