@@ -53,6 +53,8 @@ def main():
     from mgproto_amd.parallel import Comm, BucketedGradReducer, make_dp_correct
     from mgproto_amd.engine.trainer import EMRunner
 
+    from mgproto_amd.utils.helpers import setup_miopen_db
+    setup_miopen_db()
     # MIOpen exhaustive find for the (static) conv shapes: the search cost
     # lands in warmup; without it MIOpen's fallback solvers dominate the step
     # (conv bwd-weight measured at 74% of GPU time under FIND_MODE=FAST)

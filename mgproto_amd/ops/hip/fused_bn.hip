@@ -44,43 +44,70 @@ __device__ __forceinline__ short f2bf(float f) {
 // forward: per-channel sums (training stats)
 // ---------------------------------------------------------------------------
 
-// Each block covers rows_per_iter = 256/(C/8) rows per iteration (all 256
-// threads active for any C), accumulates per-thread partials in registers,
-// combines row-groups through LDS atomics, and emits ONE global atomic per
-// channel per block.  Wave lanes read consecutive rows' consecutive 16-byte
-// chunks -> fully coalesced.  For C/8 >= 256 the channel axis alone fills
-// the block (grid.y splits channels).
+// Two-stage deterministic reduction (no atomics): stage 1 blocks cover
+// rows_per_iter = 256/(C/8) rows per iteration (all 256 threads active for
+// any C), accumulate per-thread partials in registers, tree-combine
+// row-groups through LDS, and write ONE partial row [2C] per block; a tiny
+// stage-2 kernel sums the partial rows.  Wave lanes read consecutive rows'
+// consecutive 16-byte chunks -> fully coalesced.
+
+// LDS combine helper: each thread holds 16 floats (8 s, 8 q) for channel
+// chunk (tid % tpr); writes lds[tid][16] then 256 threads re-reduce the
+// rpi row-groups serially per output element. Emits block partial [2C].
+__device__ __forceinline__
+void bn_block_partial(float (&s)[8], float (&q)[8], float* lds, int tpr,
+                      float* __restrict__ partial, int C) {
+    const int tid = threadIdx.x;
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+        lds[tid * 16 + i] = s[i];
+        lds[tid * 16 + 8 + i] = q[i];
+    }
+    __syncthreads();
+    const int rpi = 256 / tpr;
+    for (int e = tid; e < tpr * 16; e += 256) {
+        const int chunk = e / 16;
+        const int comp = e % 16;
+        float acc = 0.f;
+        for (int r = 0; r < rpi; ++r)
+            acc += lds[(r * tpr + chunk) * 16 + comp];
+        const int c = chunk * 8 + (comp & 7);
+        partial[(comp < 8 ? c : C + c)] = acc;
+    }
+}
+
 __global__ __launch_bounds__(256)
-void bn_stats_kernel(const short* __restrict__ x, float* __restrict__ sums,
+void bn_stats_kernel(const short* __restrict__ x,
+                     float* __restrict__ partials,  // [gridDim.x, 2C]
                      long M, int C) {
-    extern __shared__ __attribute__((aligned(16))) float lds[];  // [2C]
+    __shared__ __attribute__((aligned(16))) float lds[256 * 16];
     const int tpr = C / 8;
-    if (tpr >= 256) {                       // channel-split path, no sharing
+    float s[8] = {0}, q[8] = {0};
+    if (tpr >= 256) {                       // channel-split path
         const int c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
-        if (c8 >= C) return;
-        float s[8] = {0}, q[8] = {0};
-        for (long m = blockIdx.x; m < M; m += gridDim.x) {
-            const short8 v = *reinterpret_cast<const short8*>(x + m * C + c8);
+        float* partial = partials + ((long)blockIdx.y * gridDim.x
+                                     + blockIdx.x) * 2 * C;
+        if (c8 < C) {
+            for (long m = blockIdx.x; m < M; m += gridDim.x) {
+                const short8 v = *reinterpret_cast<const short8*>(x + m * C + c8);
+                #pragma unroll
+                for (int i = 0; i < 8; ++i) {
+                    const float f = bf2f(v[i]);
+                    s[i] += f; q[i] += f * f;
+                }
+            }
             #pragma unroll
             for (int i = 0; i < 8; ++i) {
-                const float f = bf2f(v[i]);
-                s[i] += f; q[i] += f * f;
+                partial[c8 + i] = s[i];
+                partial[C + c8 + i] = q[i];
             }
-        }
-        #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-            atomicAdd(&sums[c8 + i], s[i]);
-            atomicAdd(&sums[C + c8 + i], q[i]);
         }
         return;
     }
-    for (int i = threadIdx.x; i < 2 * C; i += 256) lds[i] = 0.f;
-    __syncthreads();
     const int rpi = 256 / tpr;
     const int rsub = threadIdx.x / tpr;
     const int c8 = (threadIdx.x % tpr) * 8;
     if (rsub < rpi) {
-        float s[8] = {0}, q[8] = {0};
         for (long m = (long)blockIdx.x * rpi + rsub; m < M;
              m += (long)gridDim.x * rpi) {
             const short8 v = *reinterpret_cast<const short8*>(x + m * C + c8);
@@ -90,15 +117,20 @@ void bn_stats_kernel(const short* __restrict__ x, float* __restrict__ sums,
                 s[i] += f; q[i] += f * f;
             }
         }
-        #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-            atomicAdd(&lds[c8 + i], s[i]);
-            atomicAdd(&lds[C + c8 + i], q[i]);
-        }
     }
-    __syncthreads();
-    for (int i = threadIdx.x; i < 2 * C; i += 256)
-        atomicAdd(&sums[i], lds[i]);
+    bn_block_partial(s, q, lds, tpr, partials + (long)blockIdx.x * 2 * C, C);
+}
+
+// stage 2: sums[e] = sum over nb partial rows
+__global__ __launch_bounds__(256)
+void bn_partial_sum_kernel(const float* __restrict__ partials,
+                           float* __restrict__ sums, int nb, int C2) {
+    const int e = blockIdx.x * 256 + threadIdx.x;
+    if (e >= C2) return;
+    float acc = 0.f;
+    for (int b = 0; b < nb; ++b)
+        acc += partials[(long)b * C2 + e];
+    sums[e] = acc;
 }
 
 // finalize: mean/rstd, running-stat update, scale/shift for the apply pass
@@ -193,47 +225,47 @@ void bn_bwd_reduce_kernel(const short* __restrict__ dy,
                           const short* __restrict__ x,
                           const float* __restrict__ save_mean,
                           const float* __restrict__ save_rstd,
-                          float* __restrict__ sums,  // [2C]
+                          float* __restrict__ partials,  // [nblocks, 2C]
                           long M, int C) {
-    extern __shared__ __attribute__((aligned(16))) float lds[];  // [2C]
+    __shared__ __attribute__((aligned(16))) float lds[256 * 16];
     const int tpr = C / 8;
+    float sd[8] = {0}, sx[8] = {0};
     if (tpr >= 256) {
         const int c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
-        if (c8 >= C) return;
-        float sd[8] = {0}, sx[8] = {0};
-        float mean[8], rstd[8];
-        #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-            mean[i] = save_mean[c8 + i];
-            rstd[i] = save_rstd[c8 + i];
-        }
-        for (long m = blockIdx.x; m < M; m += gridDim.x) {
-            const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
-            const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
-            short8 yv;
-            if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+        float* partial = partials + ((long)blockIdx.y * gridDim.x
+                                     + blockIdx.x) * 2 * C;
+        if (c8 < C) {
+            float mean[8], rstd[8];
             #pragma unroll
             for (int i = 0; i < 8; ++i) {
-                float gf = bf2f(g[i]);
-                if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
-                const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
-                sd[i] += gf; sx[i] += gf * xhat;
+                mean[i] = save_mean[c8 + i];
+                rstd[i] = save_rstd[c8 + i];
             }
-        }
-        #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-            atomicAdd(&sums[c8 + i], sd[i]);
-            atomicAdd(&sums[C + c8 + i], sx[i]);
+            for (long m = blockIdx.x; m < M; m += gridDim.x) {
+                const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
+                const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
+                short8 yv;
+                if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+                #pragma unroll
+                for (int i = 0; i < 8; ++i) {
+                    float gf = bf2f(g[i]);
+                    if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
+                    const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
+                    sd[i] += gf; sx[i] += gf * xhat;
+                }
+            }
+            #pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                partial[c8 + i] = sd[i];
+                partial[C + c8 + i] = sx[i];
+            }
         }
         return;
     }
-    for (int i = threadIdx.x; i < 2 * C; i += 256) lds[i] = 0.f;
-    __syncthreads();
     const int rpi = 256 / tpr;
     const int rsub = threadIdx.x / tpr;
     const int c8 = (threadIdx.x % tpr) * 8;
     if (rsub < rpi) {
-        float sd[8] = {0}, sx[8] = {0};
         float mean[8], rstd[8];
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
@@ -254,15 +286,8 @@ void bn_bwd_reduce_kernel(const short* __restrict__ dy,
                 sd[i] += gf; sx[i] += gf * xhat;
             }
         }
-        #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-            atomicAdd(&lds[c8 + i], sd[i]);
-            atomicAdd(&lds[C + c8 + i], sx[i]);
-        }
     }
-    __syncthreads();
-    for (int i = threadIdx.x; i < 2 * C; i += 256)
-        atomicAdd(&sums[i], lds[i]);
+    bn_block_partial(sd, sx, lds, tpr, partials + (long)blockIdx.x * 2 * C, C);
 }
 
 // apply: dx = gamma*rstd * (dy_eff - sum_dy/M - xhat*sum_dyxhat/M)
@@ -365,18 +390,27 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
 
     const int tpr = C / 8;
     const int grid_y = tpr >= 256 ? cdiv(tpr, 256) : 1;
-    const int lds_bytes = 2 * C * sizeof(float);
     auto reduce_grid = [&]() {
         if (tpr >= 256)
-            return dim3(std::max(1, std::min((int)M, 4096 / grid_y)), grid_y);
+            return dim3(std::max(1, std::min((int)M, 64)), grid_y);
         const int rpi = 256 / tpr;
         return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 1280)), 1);
     };
     if (training) {
-        auto sums = torch::zeros({2 * C}, fopt);
-        hipLaunchKernelGGL(bn_stats_kernel, reduce_grid(), dim3(256),
-                           lds_bytes, stream, (const short*)x.data_ptr(),
-                           sums.data_ptr<float>(), M, C);
+        const dim3 rgrid = reduce_grid();
+        const int nb = rgrid.x * rgrid.y;
+        // zero-filled for the channel-split path (each block writes only
+        // its channel slice of its partial row)
+        auto partials = (tpr >= 256)
+            ? torch::zeros({nb, 2 * C}, fopt)
+            : torch::empty({nb, 2 * C}, fopt);
+        auto sums = torch::empty({2 * C}, fopt);
+        hipLaunchKernelGGL(bn_stats_kernel, rgrid, dim3(256),
+                           0, stream, (const short*)x.data_ptr(),
+                           partials.data_ptr<float>(), M, C);
+        hipLaunchKernelGGL(bn_partial_sum_kernel, dim3(cdiv(2 * C, 256)),
+                           dim3(256), 0, stream, partials.data_ptr<float>(),
+                           sums.data_ptr<float>(), nb, 2 * C);
         hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256),
                            0, stream, sums.data_ptr<float>(),
                            weight.data_ptr<float>(), bias.data_ptr<float>(),
@@ -433,35 +467,42 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
 
     const int tpr = C / 8;
     const int grid_y = tpr >= 256 ? cdiv(tpr, 256) : 1;
-    const int lds_bytes = 2 * C * sizeof(float);
     auto reduce_grid = [&]() {
         if (tpr >= 256)
-            return dim3(std::max(1, std::min((int)M, 4096 / grid_y)), grid_y);
+            return dim3(std::max(1, std::min((int)M, 64)), grid_y);
         const int rpi = 256 / tpr;
         return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 1280)), 1);
     };
+    const dim3 rgrid = reduce_grid();
+    const int nb = rgrid.x * rgrid.y;
+    auto partials = (tpr >= 256)
+        ? torch::zeros({nb, 2 * C}, fopt)
+        : torch::empty({nb, 2 * C}, fopt);
     const long total = M * (C / 8);
     const int blocks = (int)std::min<long>(cdiv(total, 256), 2048);
 
     if (training) {
         if (relu)
             hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>),
-                               reduce_grid(), dim3(256), lds_bytes, stream,
+                               rgrid, dim3(256), 0, stream,
                                (const short*)dy.data_ptr(),
                                (const short*)y.data_ptr(),
                                (const short*)x.data_ptr(),
                                save_mean.data_ptr<float>(),
                                save_rstd.data_ptr<float>(),
-                               sums.data_ptr<float>(), M, C);
+                               partials.data_ptr<float>(), M, C);
         else
             hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>),
-                               reduce_grid(), dim3(256), lds_bytes, stream,
+                               rgrid, dim3(256), 0, stream,
                                (const short*)dy.data_ptr(),
                                (const short*)y.data_ptr(),
                                (const short*)x.data_ptr(),
                                save_mean.data_ptr<float>(),
                                save_rstd.data_ptr<float>(),
-                               sums.data_ptr<float>(), M, C);
+                               partials.data_ptr<float>(), M, C);
+        hipLaunchKernelGGL(bn_partial_sum_kernel, dim3(cdiv(2 * C, 256)),
+                           dim3(256), 0, stream, partials.data_ptr<float>(),
+                           sums.data_ptr<float>(), nb, 2 * C);
         #define BWD_APPLY(RELU_, RES_) \
             hipLaunchKernelGGL((bn_bwd_apply_kernel<RELU_, RES_>), \
                                dim3(blocks), dim3(256), 0, stream, \
@@ -484,22 +525,25 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
         // reduce (xhat uses running stats)
         if (relu)
             hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>),
-                               reduce_grid(), dim3(256), lds_bytes, stream,
+                               rgrid, dim3(256), 0, stream,
                                (const short*)dy.data_ptr(),
                                (const short*)y.data_ptr(),
                                (const short*)x.data_ptr(),
                                save_mean.data_ptr<float>(),
                                save_rstd.data_ptr<float>(),
-                               sums.data_ptr<float>(), M, C);
+                               partials.data_ptr<float>(), M, C);
         else
             hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>),
-                               reduce_grid(), dim3(256), lds_bytes, stream,
+                               rgrid, dim3(256), 0, stream,
                                (const short*)dy.data_ptr(),
                                (const short*)y.data_ptr(),
                                (const short*)x.data_ptr(),
                                save_mean.data_ptr<float>(),
                                save_rstd.data_ptr<float>(),
-                               sums.data_ptr<float>(), M, C);
+                               partials.data_ptr<float>(), M, C);
+        hipLaunchKernelGGL(bn_partial_sum_kernel, dim3(cdiv(2 * C, 256)),
+                           dim3(256), 0, stream, partials.data_ptr<float>(),
+                           sums.data_ptr<float>(), nb, 2 * C);
         #define EVAL_APPLY(RELU_, RES_) \
             hipLaunchKernelGGL((bn_bwd_eval_kernel<RELU_, RES_>), \
                                dim3(blocks), dim3(256), 0, stream, \

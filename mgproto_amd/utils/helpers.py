@@ -58,3 +58,28 @@ def find_high_activation_crop(activation_map: np.ndarray, percentile: float = 95
     lower_y, upper_y = int(ys.min()), int(ys.max())
     lower_x, upper_x = int(xs.min()), int(xs.max())
     return (lower_y, upper_y + 1, lower_x, upper_x + 1)
+
+
+def setup_miopen_db():
+    """Seed MIOpen's user find-db from the in-repo copy (gfx950).
+
+    MIOpen's exhaustive find (torch.backends.cudnn.benchmark) costs ~2-3
+    minutes per fresh box; the resulting tuning db is tiny text, so we ship
+    it and point MIOPEN_USER_DB_PATH at a writable copy. No-op if the env
+    var is already set or the db directory is absent.
+    """
+    import shutil
+    import tempfile
+    if os.environ.get('MIOPEN_USER_DB_PATH'):
+        return
+    src = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                       'miopen_db')
+    if not os.path.isdir(src) or not os.listdir(src):
+        return
+    dst = os.path.join(tempfile.gettempdir(), 'mgproto_miopen_db')
+    os.makedirs(dst, exist_ok=True)
+    for f in os.listdir(src):
+        tgt = os.path.join(dst, f)
+        if not os.path.exists(tgt):
+            shutil.copy2(os.path.join(src, f), tgt)
+    os.environ['MIOPEN_USER_DB_PATH'] = dst
