@@ -37,6 +37,8 @@ def try_capture(name, fn, warm=3):
 
 
 def main():
+    from mgproto_amd.utils.helpers import setup_miopen_db
+    setup_miopen_db()
     torch.backends.cudnn.benchmark = True
     dev = torch.device('cuda', 0)
     C, K, d = 200, 10, 64

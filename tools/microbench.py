@@ -37,6 +37,8 @@ def main():
     ap.add_argument('--no-find', action='store_true')
     args = ap.parse_args()
 
+    from mgproto_amd.utils.helpers import setup_miopen_db
+    setup_miopen_db()
     torch.backends.cudnn.benchmark = not args.no_find
     dev = torch.device('cuda', 0)
     C, K, d = 200, 10, 64

@@ -105,6 +105,8 @@ def main():
                     else torch.device('cpu')))
     if device.type == 'cuda':
         torch.cuda.set_device(device)
+        from mgproto_amd.utils.helpers import setup_miopen_db
+        setup_miopen_db()
         torch.backends.cudnn.benchmark = True
 
     model_dir = args.out or 'saved_models/{}/{}/'.format(args.arch, datestr())
