@@ -127,10 +127,18 @@ void bn_partial_sum_kernel(const float* __restrict__ partials,
                            float* __restrict__ sums, int nb, int C2) {
     const int e = blockIdx.x * 256 + threadIdx.x;
     if (e >= C2) return;
-    float acc = 0.f;
-    for (int b = 0; b < nb; ++b)
-        acc += partials[(long)b * C2 + e];
-    sums[e] = acc;
+    // 8 independent accumulators keep 8 loads in flight (the serial form
+    // was 1280 dependent L2 round trips = ~270 us per call)
+    float a[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    int b = 0;
+    for (; b + 8 <= nb; b += 8) {
+        #pragma unroll
+        for (int i = 0; i < 8; ++i)
+            a[i] += partials[(long)(b + i) * C2 + e];
+    }
+    for (; b < nb; ++b)
+        a[0] += partials[(long)b * C2 + e];
+    sums[e] = ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
 }
 
 // finalize: mean/rstd, running-stat update, scale/shift for the apply pass
@@ -185,25 +193,41 @@ void bn_eval_coeffs_kernel(const float* __restrict__ weight,
 }
 
 // apply: y = [relu](x*scale + shift [+ res]); RELU/RES are compile-time
+// channel-resident thread mapping (as bn_stats): fixed c8 per thread, row
+// loop — no per-element integer division, coalesced 16B lanes.
 template <bool RELU, bool RES>
 __global__ __launch_bounds__(256)
 void bn_apply_kernel(const short* __restrict__ x,
                      const short* __restrict__ res,
                      const float* __restrict__ scale_shift,
                      short* __restrict__ y, long M, int C) {
-    const long total = M * (C / 8);
-    for (long t = (long)blockIdx.x * 256 + threadIdx.x; t < total;
-         t += (long)gridDim.x * 256) {
-        const long m = t / (C / 8);
-        const int c8 = (int)(t % (C / 8)) * 8;
+    const int tpr = C / 8;
+    int c8, rsub, rpi;
+    if (tpr >= 256) {
+        c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
+        if (c8 >= C) return;
+        rsub = 0; rpi = 1;
+    } else {
+        rpi = 256 / tpr;
+        rsub = threadIdx.x / tpr;
+        c8 = (threadIdx.x % tpr) * 8;
+        if (rsub >= rpi) return;
+    }
+    float sc[8], sh[8];
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+        sc[i] = scale_shift[c8 + i];
+        sh[i] = scale_shift[C + c8 + i];
+    }
+    for (long m = (long)blockIdx.x * rpi + rsub; m < M;
+         m += (long)gridDim.x * rpi) {
         const short8 v = *reinterpret_cast<const short8*>(x + m * C + c8);
         short8 r;
         if (RES) r = *reinterpret_cast<const short8*>(res + m * C + c8);
         short8 o;
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
-            float f = fmaf(bf2f(v[i]), scale_shift[c8 + i],
-                           scale_shift[C + c8 + i]);
+            float f = fmaf(bf2f(v[i]), sc[i], sh[i]);
             if (RES) f += bf2f(r[i]);
             if (RELU) f = fmaxf(f, 0.f);
             o[i] = f2bf(f);
@@ -305,11 +329,29 @@ void bn_bwd_apply_kernel(const short* __restrict__ dy,
                          short* __restrict__ dres,
                          long M, int C) {
     const float invM = 1.f / (float)M;
-    const long total = M * (C / 8);
-    for (long t = (long)blockIdx.x * 256 + threadIdx.x; t < total;
-         t += (long)gridDim.x * 256) {
-        const long m = t / (C / 8);
-        const int c8 = (int)(t % (C / 8)) * 8;
+    const int tpr = C / 8;
+    int c8, rsub, rpi;
+    if (tpr >= 256) {
+        c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
+        if (c8 >= C) return;
+        rsub = 0; rpi = 1;
+    } else {
+        rpi = 256 / tpr;
+        rsub = threadIdx.x / tpr;
+        c8 = (threadIdx.x % tpr) * 8;
+        if (rsub >= rpi) return;
+    }
+    float mean[8], rstd[8], gw[8], sdy[8], sxh[8];
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+        mean[i] = save_mean[c8 + i];
+        rstd[i] = save_rstd[c8 + i];
+        gw[i] = weight[c8 + i] * rstd[i];
+        sdy[i] = sums[c8 + i] * invM;
+        sxh[i] = sums[C + c8 + i] * invM;
+    }
+    for (long m = (long)blockIdx.x * rpi + rsub; m < M;
+         m += (long)gridDim.x * rpi) {
         const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
         const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
         short8 yv;
@@ -317,13 +359,11 @@ void bn_bwd_apply_kernel(const short* __restrict__ dy,
         short8 odx, odr;
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
-            const int c = c8 + i;
             float gf = bf2f(g[i]);
             if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
             if (RES) odr[i] = f2bf(gf);
-            const float xhat = (bf2f(xv[i]) - save_mean[c]) * save_rstd[c];
-            const float t1 = gf - sums[c] * invM - xhat * sums[C + c] * invM;
-            odx[i] = f2bf(weight[c] * save_rstd[c] * t1);
+            const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
+            odx[i] = f2bf(gw[i] * (gf - sdy[i] - xhat * sxh[i]));
         }
         *reinterpret_cast<short8*>(dx + m * C + c8) = odx;
         if (RES) *reinterpret_cast<short8*>(dres + m * C + c8) = odr;
@@ -341,22 +381,34 @@ void bn_bwd_eval_kernel(const short* __restrict__ dy,
                         short* __restrict__ dx,
                         short* __restrict__ dres,
                         long M, int C) {
-    const long total = M * (C / 8);
-    for (long t = (long)blockIdx.x * 256 + threadIdx.x; t < total;
-         t += (long)gridDim.x * 256) {
-        const long m = t / (C / 8);
-        const int c8 = (int)(t % (C / 8)) * 8;
+    const int tpr = C / 8;
+    int c8, rsub, rpi;
+    if (tpr >= 256) {
+        c8 = (blockIdx.y * 256 + threadIdx.x) * 8;
+        if (c8 >= C) return;
+        rsub = 0; rpi = 1;
+    } else {
+        rpi = 256 / tpr;
+        rsub = threadIdx.x / tpr;
+        c8 = (threadIdx.x % tpr) * 8;
+        if (rsub >= rpi) return;
+    }
+    float gw[8];
+    #pragma unroll
+    for (int i = 0; i < 8; ++i)
+        gw[i] = weight[c8 + i] * save_rstd[c8 + i];
+    for (long m = (long)blockIdx.x * rpi + rsub; m < M;
+         m += (long)gridDim.x * rpi) {
         const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
         short8 yv;
         if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
         short8 odx, odr;
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
-            const int c = c8 + i;
             float gf = bf2f(g[i]);
             if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
             if (RES) odr[i] = f2bf(gf);
-            odx[i] = f2bf(weight[c] * save_rstd[c] * gf);
+            odx[i] = f2bf(gw[i] * gf);
         }
         *reinterpret_cast<short8*>(dx + m * C + c8) = odx;
         if (RES) *reinterpret_cast<short8*>(dres + m * C + c8) = odr;
@@ -394,7 +446,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
         if (tpr >= 256)
             return dim3(std::max(1, std::min((int)M, 64)), grid_y);
         const int rpi = 256 / tpr;
-        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 1280)), 1);
+        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 256)), 1);
     };
     if (training) {
         const dim3 rgrid = reduce_grid();
@@ -431,15 +483,16 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
                            scale_shift.data_ptr<float>(), C, (float)eps);
     }
 
-    const long total = M * (C / 8);
-    const int blocks = (int)std::min<long>(cdiv(total, 256), 2048);
+    const dim3 agrid = (tpr >= 256)
+        ? dim3(std::max(1, std::min((int)M, 1024 / grid_y)), grid_y)
+        : dim3((int)std::max<long>(1, std::min<long>(cdiv(M, 256 / tpr), 1024)), 1);
     const short* res_ptr = nullptr;
     if (residual.has_value()) {
         CHECK_BN(residual.value());
         res_ptr = (const short*)residual.value().data_ptr();
     }
     #define APPLY(RELU_, RES_) \
-        hipLaunchKernelGGL((bn_apply_kernel<RELU_, RES_>), dim3(blocks), \
+        hipLaunchKernelGGL((bn_apply_kernel<RELU_, RES_>), agrid, \
                            dim3(256), 0, stream, (const short*)x.data_ptr(), \
                            res_ptr, scale_shift.data_ptr<float>(), \
                            (short*)y.data_ptr(), M, C)
@@ -471,15 +524,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
         if (tpr >= 256)
             return dim3(std::max(1, std::min((int)M, 64)), grid_y);
         const int rpi = 256 / tpr;
-        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 1280)), 1);
+        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 256)), 1);
     };
     const dim3 rgrid = reduce_grid();
     const int nb = rgrid.x * rgrid.y;
     auto partials = (tpr >= 256)
         ? torch::zeros({nb, 2 * C}, fopt)
         : torch::empty({nb, 2 * C}, fopt);
-    const long total = M * (C / 8);
-    const int blocks = (int)std::min<long>(cdiv(total, 256), 2048);
+    const dim3 agrid = (tpr >= 256)
+        ? dim3(std::max(1, std::min((int)M, 1024 / grid_y)), grid_y)
+        : dim3((int)std::max<long>(1, std::min<long>(cdiv(M, 256 / tpr), 1024)), 1);
 
     if (training) {
         if (relu)
@@ -505,7 +559,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                            sums.data_ptr<float>(), nb, 2 * C);
         #define BWD_APPLY(RELU_, RES_) \
             hipLaunchKernelGGL((bn_bwd_apply_kernel<RELU_, RES_>), \
-                               dim3(blocks), dim3(256), 0, stream, \
+                               agrid, dim3(256), 0, stream, \
                                (const short*)dy.data_ptr(), \
                                (const short*)y.data_ptr(), \
                                (const short*)x.data_ptr(), \
@@ -546,7 +600,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                            sums.data_ptr<float>(), nb, 2 * C);
         #define EVAL_APPLY(RELU_, RES_) \
             hipLaunchKernelGGL((bn_bwd_eval_kernel<RELU_, RES_>), \
-                               dim3(blocks), dim3(256), 0, stream, \
+                               agrid, dim3(256), 0, stream, \
                                (const short*)dy.data_ptr(), \
                                (const short*)y.data_ptr(), \
                                save_rstd.data_ptr<float>(), \
