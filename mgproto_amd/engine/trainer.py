@@ -81,6 +81,12 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
     total_aux = torch.zeros((), device=device)
     n_batches = 0
 
+    # host-side mirrors of the EM gating state: the reference re-reads
+    # device counters every batch (train_and_test.py:61-62), which would be
+    # one host sync per step; sync once per epoch instead
+    mem_nonempty = bool(int(m.queue.mem_len.sum()) > 0) if update_GMM else False
+    iter_base = int(m.iteration_counter)
+
     for i, batch in enumerate(dataloader):
         image, label = batch[0], batch[1]
         image = image.to(device, non_blocking=True)
@@ -117,8 +123,9 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
         optimizer.step()
 
         # EM update (reference train_and_test.py:61-63; update_interval=1)
-        if update_GMM and int(m.queue.mem_len.sum()) > 0:
-            if int(m.iteration_counter) % m.update_interval == 0:
+        if update_GMM:
+            mem_nonempty = mem_nonempty or (i == 0 and target.numel() > 0)
+            if mem_nonempty and (iter_base + i + 1) % m.update_interval == 0:
                 (em_runner.run() if em_runner is not None else m.update_GMM())
 
         if print_every and i % print_every == 0:

@@ -46,7 +46,7 @@ def build_loaders(cfg: Settings, comm):
     if os.path.isdir(cfg.train_dir):
         return L.build_image_loaders(cfg, world=world, rank=rank)
     # synthetic fallback (no datasets in this environment)
-    n = 1600
+    n = 32 if os.environ.get('MGPROTO_TINY_TEST') == '1' else 1600
     train_ds = SyntheticImages(n, cfg.num_classes, cfg.img_size)
     push_ds = SyntheticImages(n, cfg.num_classes, cfg.img_size, normalize=False)
     test_ds = SyntheticImages(n // 4, cfg.num_classes, cfg.img_size, seed=1)
@@ -94,6 +94,19 @@ def main():
                    mine_K=args.mine_level)
     if args.addon:
         cfg.add_on_layers_type = args.addon
+    if os.environ.get('MGPROTO_TINY_TEST') == '1':
+        # CI-sized config: small synthetic problem, CPU-friendly
+        cfg.img_size = 64
+        cfg.num_classes = 10
+        cfg.prototype_shape = (30, 32, 1, 1)
+        cfg.add_on_layers_type = 'regular'
+        cfg.train_batch_size = cfg.test_batch_size = 8
+        cfg.train_push_batch_size = 8
+        cfg.num_workers = 0
+        cfg.mine_start = 0
+        cfg.updateGMM_start = 0
+        cfg.push_start = 0
+        cfg.push_epochs = [0]
     if args.epochs is not None:
         cfg.num_train_epochs = args.epochs
         cfg.push_epochs = [i for i in range(args.epochs) if i % 10 == 0]
@@ -229,7 +242,7 @@ def main():
                                        accu, 0.00, log=log)
 
     # final pruning (reference main.py:285)
-    ppnet.prune_prototypes_topM(top_M=8)
+    ppnet.prune_prototypes_topM(top_M=min(8, ppnet.num_prototypes_per_class))
     accu, _ = tnt.test(ppnet, (test_loader,), log=log, device=device,
                        amp_dtype=cfg.amp_dtype, comm=comm)
     if rank == 0:
