@@ -42,6 +42,28 @@ def main():
                     help='disable hipGraph capture of the training step')
     ap.add_argument('--eager', action='store_true',
                     help='allow the PyTorch fallback for the prototype ops')
+    ap.add_argument('--preset', type=str, default=None,
+                    choices=['cub-r50', 'cars-densenet161', 'ood-resnet152',
+                             'pets-vgg19'],
+                    help='BASELINE.json config presets (2-5); flags still '
+                         'override')
+    # presets (BASELINE.json configs): apply before final parse so explicit
+    # flags win
+    import sys as _sys
+    pre, _ = ap.parse_known_args()
+    presets = {
+        'cars-densenet161': dict(arch='densenet161', classes=196,
+                                 addon='regular'),
+        'ood-resnet152': dict(arch='resnet152', classes=200, addon='regular'),
+        'pets-vgg19': dict(arch='vgg19', classes=37, batch=256, mem=4000,
+                           addon='regular'),
+    }
+    if pre.preset in presets:
+        explicit = {a.lstrip('-').replace('-', '_').split('=')[0]
+                    for a in _sys.argv[1:] if a.startswith('--')}
+        for k, v in presets[pre.preset].items():
+            if k not in explicit:
+                ap.set_defaults(**{k: v})
     args = ap.parse_args()
 
     if args.eager:

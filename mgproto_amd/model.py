@@ -381,6 +381,43 @@ class MGProto(nn.Module):
         stepped = means - self.prototype_lr * (m / bc1) / denom
         return torch.where(act3, stepped, means)
 
+    # ------------------------------------------- EM API parity (per class)
+    # Single-class forms matching the reference's method surface
+    # (model.py:303-427); update_GMM uses the batched ops versions.
+
+    def _check_size(self, x):
+        if len(x.size()) == 2:
+            x = x.unsqueeze(1)
+        return x
+
+    def _estimate_log_prob(self, x, mu, var, eps=1e-10):
+        """[n, 1, d] x [1, k, d] -> [n, k, 1] (reference model.py:323-336)."""
+        x = self._check_size(x)
+        d = x.shape[-1]
+        log_p = ((x - mu) / (var + eps)).pow(2).sum(dim=2, keepdim=True)
+        log_sigma = torch.log(var + eps).sum(dim=2, keepdim=True)
+        return -0.5 * d * math.log(2 * math.pi) - log_sigma - 0.5 * log_p
+
+    def _e_step(self, x, mu, var, pi, eps=1e-10):
+        """(mean log-evidence, log-responsibilities) — reference :303-321."""
+        x = self._check_size(x)
+        wlp = self._estimate_log_prob(x, mu, var) + torch.log(pi + eps)
+        log_norm = torch.logsumexp(wlp, dim=1, keepdim=True)
+        return torch.mean(log_norm), wlp - log_norm
+
+    def _m_step(self, x, log_resp, eps=1e-10):
+        """Closed-form M-step (reference :338-365 — dead code there, kept
+        for API parity; update_GMM uses _m_step-diversified semantics)."""
+        x = self._check_size(x)
+        resp = torch.exp(log_resp)
+        resp = (resp + self.alpha) / (resp + self.alpha).sum(1, keepdim=True)
+        pi = torch.sum(resp, dim=0, keepdim=True) + eps
+        mu = torch.sum(resp * x, dim=0, keepdim=True) / pi
+        x2 = (resp * x * x).sum(0, keepdim=True) / pi
+        xmu = (resp * mu * x).sum(0, keepdim=True) / pi
+        var = (x2 - 2 * xmu + mu * mu + eps).sqrt()
+        return pi / x.shape[0], mu, var
+
     # --------------------------------------------------------------- scoring
     def _score(self, x, mu, var, pi, as_average=True, eps=1e-10):
         """Data log-likelihood under one class's mixture (reference :403-421)."""

@@ -137,3 +137,18 @@ def vgg19_features(pretrained=False, **kwargs):
 
 def vgg19_bn_features(pretrained=False, **kwargs):
     return _vgg('vgg19_bn', 'E', True, pretrained, **kwargs)
+
+
+class VGG_vanilla(nn.Module):
+    """Plain VGG-19 classifier head over the full trunk (reference
+    vgg_features.py:110-124); used for baseline comparisons only."""
+
+    def __init__(self, num_classes=200, pretrained=False):
+        super().__init__()
+        self.vgg19_f = vgg19_features(pretrained=pretrained,
+                                      final_maxpool=True, final_relu=True)
+        self.addons = nn.Linear(512 * 7 * 7, num_classes)
+
+    def forward(self, x):
+        x = self.vgg19_f(x)
+        return self.addons(x.flatten(1))

@@ -161,6 +161,40 @@ class MemoryBank(nn.Module):
             return None, None
         return torch.cat(out_data, 0), torch.cat(out_label, 0)
 
+    @torch.no_grad()
+    def pull_fix_length(self, label: torch.Tensor):
+        """Reference memory.py:95-133: per query row, a fixed-length sample
+        of features whose classes are flagged in that row's one-hot."""
+        assert label.dim() == 2 and label.size(1) == self.num_classes
+        pull_num = self.cap_cls * self.fix_length_mult
+        label = label.detach()
+        indices = torch.nonzero(label.max(0)[0], as_tuple=False).flatten()
+        cand_data, cand_label = [], []
+        for i in indices.tolist():
+            L = int(self.mem_len[i])
+            if L == 0:
+                continue
+            cand_data.append(self._logical(i))
+            cand_label.append(torch.full((L,), i, dtype=torch.int64,
+                                         device=self.mem.device))
+        if not cand_data:
+            return None, None
+        cand_data = torch.cat(cand_data, 0)
+        cand_label = torch.cat(cand_label, 0)
+        oh = torch.nn.functional.one_hot(cand_label, self.num_classes).float()
+        cand_mask = torch.matmul(label.float(), oh.T) > 0.999
+        if cand_mask.sum(1).min() < 1:
+            return None, None
+        out = []
+        for i in range(label.size(0)):
+            data = cand_data[cand_mask[i]]
+            while data.size(0) < pull_num:
+                data = data.repeat(2, 1)
+            if data.size(0) > pull_num:
+                data = data[torch.randperm(data.size(0))[:pull_num]]
+            out.append(data.unsqueeze(1))
+        return torch.cat(out, 1).contiguous(), None
+
     def pull(self, *args, **kwargs):
         if self.mode == 'all':
             return self.pull_all(*args, **kwargs)
