@@ -30,14 +30,17 @@ def _ref_bn(x, bn, relu, residual, training):
     return F.relu(y) if relu else y
 
 
-@pytest.mark.parametrize('relu,res,training',
-                         [(False, False, True), (True, False, True),
-                          (True, True, True), (False, True, True),
-                          (True, False, False), (False, False, False)])
-def test_bn_forward_parity(relu, res, training):
+@pytest.mark.parametrize('relu,res,training,C',
+                         [(False, False, True, 64), (True, False, True, 64),
+                          (True, True, True, 64), (False, True, True, 64),
+                          (True, False, False, 64), (False, False, False, 64),
+                          # DenseNet-like channel counts: C=96 -> 256%tpr!=0
+                          # row groups; C=2208 -> channel-split (tpr>=256)
+                          (True, False, True, 96), (True, False, True, 2208),
+                          (True, False, True, 288), (False, False, True, 2048)])
+def test_bn_forward_parity(relu, res, training, C):
     from mgproto_amd.models.fused_bn import bn_act
     torch.manual_seed(0)
-    C = 64
     bn = nn.BatchNorm2d(C).cuda()
     bn.weight.data.uniform_(0.5, 1.5)
     bn.bias.data.uniform_(-0.5, 0.5)
@@ -45,7 +48,7 @@ def test_bn_forward_parity(relu, res, training):
     bn.running_var.data.uniform_(0.8, 1.2)
     bn.train(training)
 
-    x, r = _pair(res=res)
+    x, r = _pair(C=C, res=res)
     rm0, rv0 = bn.running_mean.clone(), bn.running_var.clone()
     want = _ref_bn(x, bn, relu, r, training)
 
@@ -67,19 +70,19 @@ def test_bn_forward_parity(relu, res, training):
         assert torch.allclose(bn.running_var, want_rv, atol=1e-2, rtol=1e-2)
 
 
-@pytest.mark.parametrize('relu,res', [(False, False), (True, False),
-                                      (True, True)])
-def test_bn_backward_parity(relu, res):
+@pytest.mark.parametrize('relu,res,C', [(False, False, 64), (True, False, 64),
+                                        (True, True, 64), (True, False, 96),
+                                        (True, False, 2208)])
+def test_bn_backward_parity(relu, res, C):
     from mgproto_amd.models.fused_bn import bn_act
     torch.manual_seed(1)
-    C = 64
     bn1 = nn.BatchNorm2d(C).cuda().train()
     bn1.weight.data.uniform_(0.5, 1.5)
     bn1.bias.data.uniform_(-0.5, 0.5)
     bn2 = nn.BatchNorm2d(C).cuda().train()
     bn2.load_state_dict(bn1.state_dict())
 
-    x, r = _pair(seed=2, res=res)
+    x, r = _pair(C=C, seed=2, res=res)
     x1 = x.clone().requires_grad_(True)
     r1 = r.clone().requires_grad_(True) if res else None
     y1 = bn_act(x1, bn1, relu=relu, residual=r1)
