@@ -35,6 +35,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from . import ops
+from .models.conv1x1 import GemmConv2d
 from .models import (resnet18_features, resnet34_features, resnet50_features,
                      resnet101_features, resnet152_features,
                      densenet121_features, densenet161_features,
@@ -149,9 +150,9 @@ class MGProto(nn.Module):
             current_in = first_add_on_layer_in_channels
             while (current_in > d) or (len(add_on_layers) == 0):
                 current_out = max(d, (current_in // 2))
-                add_on_layers.append(nn.Conv2d(current_in, current_out, kernel_size=1))
+                add_on_layers.append(GemmConv2d(current_in, current_out, kernel_size=1))
                 add_on_layers.append(nn.ReLU())
-                add_on_layers.append(nn.Conv2d(current_out, current_out, kernel_size=1))
+                add_on_layers.append(GemmConv2d(current_out, current_out, kernel_size=1))
                 if current_out > d:
                     add_on_layers.append(nn.ReLU())
                 else:
@@ -164,13 +165,13 @@ class MGProto(nn.Module):
             # 2x bilinear upsample -> 28x28 latent grid at 224 input
             self.add_on_layers = nn.Sequential(
                 nn.Upsample(scale_factor=2, mode='bilinear', align_corners=False),
-                nn.Conv2d(first_add_on_layer_in_channels, d, kernel_size=1),
-                nn.Conv2d(d, d, kernel_size=1),
+                GemmConv2d(first_add_on_layer_in_channels, d, kernel_size=1),
+                GemmConv2d(d, d, kernel_size=1),
             )
         else:  # 'regular'
             self.add_on_layers = nn.Sequential(
-                nn.Conv2d(first_add_on_layer_in_channels, d, kernel_size=1),
-                nn.Conv2d(d, d, kernel_size=1),
+                GemmConv2d(first_add_on_layer_in_channels, d, kernel_size=1),
+                GemmConv2d(d, d, kernel_size=1),
             )
 
         self.gap = nn.AdaptiveAvgPool2d(1)

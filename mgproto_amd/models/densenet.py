@@ -22,6 +22,7 @@ import torch.nn.functional as F
 
 from .resnet import PRETRAINED_DIR
 from .fused_bn import FusedBatchNorm2d
+from .conv1x1 import GemmConv2d
 
 
 class _DenseLayer(nn.Sequential):
@@ -32,8 +33,10 @@ class _DenseLayer(nn.Sequential):
         self.add_module('norm1', FusedBatchNorm2d(num_input_features,
                                                   fused_relu=True))
         self.add_module('relu1', nn.Identity())
-        self.add_module('conv1', nn.Conv2d(num_input_features, bn_size * growth_rate,
-                                           kernel_size=1, stride=1, bias=False))
+        self.add_module('conv1', GemmConv2d(num_input_features,
+                                            bn_size * growth_rate,
+                                            kernel_size=1, stride=1,
+                                            bias=False))
         self.add_module('norm2', FusedBatchNorm2d(bn_size * growth_rate,
                                                   fused_relu=True))
         self.add_module('relu2', nn.Identity())
@@ -79,8 +82,10 @@ class _Transition(nn.Sequential):
         self.add_module('norm', FusedBatchNorm2d(num_input_features,
                                                  fused_relu=True))
         self.add_module('relu', nn.Identity())
-        self.add_module('conv', nn.Conv2d(num_input_features, num_output_features,
-                                          kernel_size=1, stride=1, bias=False))
+        self.add_module('conv', GemmConv2d(num_input_features,
+                                           num_output_features,
+                                           kernel_size=1, stride=1,
+                                           bias=False))
         self.add_module('pool', nn.AvgPool2d(kernel_size=2, stride=2))
 
     def block_conv_info(self):

@@ -29,6 +29,7 @@ import torch
 import torch.nn as nn
 
 from .fused_bn import bn_act
+from .conv1x1 import GemmConv2d
 
 PRETRAINED_DIR = os.environ.get('MGPROTO_PRETRAINED_DIR', './pretrained_models')
 
@@ -39,7 +40,9 @@ def conv3x3(in_planes, out_planes, stride=1):
 
 
 def conv1x1(in_planes, out_planes, stride=1):
-    return nn.Conv2d(in_planes, out_planes, kernel_size=1, stride=stride, bias=False)
+    # stride-1 instances take the hipBLASLt GEMM path (models/conv1x1.py)
+    return GemmConv2d(in_planes, out_planes, kernel_size=1, stride=stride,
+                      bias=False)
 
 
 class BasicBlock(nn.Module):
