@@ -20,7 +20,11 @@ import torch.nn.functional as F
 
 
 def _usable(x: torch.Tensor, conv: nn.Conv2d) -> bool:
-    if os.environ.get('MGPROTO_NO_GEMM_CONV1X1') == '1':
+    # Measured on the flagship step: plain hipBLASLt matmul on these
+    # skinny-K shapes is ~35% SLOWER than MIOpen's CK batched-GEMM picks
+    # (1236 vs 1666 img/s, profiles/bench_history_r01.md), so the GEMM
+    # path is opt-in pending TunableOp-tuned GEMMs.
+    if os.environ.get('MGPROTO_GEMM_CONV1X1') != '1':
         return False
     return (x.is_cuda and x.dim() == 4
             and x.dtype in (torch.bfloat16, torch.float16)
