@@ -12,8 +12,8 @@
 // (M = N*H*W), bf16, C % 8 == 0; statistics and parameters are fp32.
 // Each lane owns 8 consecutive channels (one 16-byte vector load), so a
 // wave reads 1 KiB contiguous — fully coalesced.  Per-channel reductions
-// accumulate block partials in registers and combine with fp32 atomics
-// (channel count is small, contention negligible).
+// are deterministic two-stage: register partials -> LDS combine -> one
+// partial row per block -> ILP-unrolled stage-2 sum (no atomics).
 
 #include <torch/extension.h>
 #include <hip/hip_runtime.h>
@@ -446,7 +446,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
         if (tpr >= 256)
             return dim3(std::max(1, std::min((int)M, 64)), grid_y);
         const int rpi = 256 / tpr;
-        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 256)), 1);
+        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 1024)), 1);
     };
     if (training) {
         const dim3 rgrid = reduce_grid();
@@ -524,7 +524,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
         if (tpr >= 256)
             return dim3(std::max(1, std::min((int)M, 64)), grid_y);
         const int rpi = 256 / tpr;
-        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 256)), 1);
+        return dim3((int)std::max<long>(1, std::min<long>(cdiv(M, rpi), 1024)), 1);
     };
     const dim3 rgrid = reduce_grid();
     const int nb = rgrid.x * rgrid.y;
