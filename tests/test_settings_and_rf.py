@@ -1,0 +1,92 @@
+"""Settings config surface + receptive-field math + losses coverage."""
+
+import math
+
+import pytest
+import torch
+
+from mgproto_amd import settings
+from mgproto_amd.utils.receptive_field import (compute_layer_rf_info,
+                                               compute_proto_layer_rf_info_v2,
+                                               compute_rf_prototype)
+
+
+def test_settings_surface_matches_reference():
+    # module constants the reference exposes (settings.py:1-52)
+    assert settings.img_size == 224
+    assert settings.num_classes == 200
+    assert settings.prototype_shape == (2000, 64, 1, 1)
+    assert settings.prototype_activation_function == 'log'
+    assert settings.add_on_layers_type == 'regular'
+    assert settings.train_batch_size == 80
+    assert settings.joint_optimizer_lrs == {'features': 1e-4,
+                                            'add_on_layers': 3e-3,
+                                            'prototype_vectors': 3e-3}
+    assert settings.warm_optimizer_lrs == {'add_on_layers': 3e-3,
+                                           'prototype_vectors': 3e-3}
+    assert settings.last_layer_optimizer_lr == 1e-4
+    assert settings.coefs == {'crs_ent': 1, 'mine': 0.2, 'aux': 0.5}
+    assert settings.num_train_epochs == 120
+    assert settings.num_warm_epochs == 0
+    assert settings.mine_start == 40
+    assert settings.updateGMM_start == 35
+    assert settings.push_start == 100
+    assert settings.push_epochs == [i for i in range(120) if i % 10 == 0]
+    for name in ('train_dir', 'test_dir', 'train_push_dir', 'test_dir_ood1',
+                 'test_dir_ood2'):
+        assert isinstance(getattr(settings, name), str)
+
+
+def test_settings_dataclass_decay_tables():
+    cfg = settings.Settings(base_architecture='resnet34')
+    assert cfg.lr_decay_epochs() == [30, 45, 60, 75, 90]
+    cfg = settings.Settings(base_architecture='resnet50')
+    assert cfg.lr_decay_epochs() == [10, 15, 20, 25, 30]
+    d = cfg.to_dict()
+    assert d['mem_capacity'] == 800 and d['mine_K'] == 20
+
+
+def test_rf_single_conv():
+    # 3x3 stride-1 pad-1 conv on 224: n stays, rf grows to 3
+    out = compute_layer_rf_info(3, 1, 1, [224, 1, 1, 0.5])
+    assert out == [224, 1, 3, 0.5]
+    # stride-2 7x7 pad-3 stem
+    out = compute_layer_rf_info(7, 2, 3, [224, 1, 1, 0.5])
+    assert out[0] == 112 and out[1] == 2 and out[2] == 7
+
+
+def test_rf_matches_backbone_grid():
+    """The RF latent grid size equals the actual backbone output size
+    (this is the conv_info fix — the reference's phantom max-pool made
+    n half the true grid)."""
+    from mgproto_amd.models import resnet34_features
+    f = resnet34_features()
+    ks, ss, ps = f.conv_info()
+    info = compute_proto_layer_rf_info_v2(224, ks, ss, ps, 1)
+    with torch.no_grad():
+        out = f(torch.randn(1, 3, 224, 224))
+    assert info[0] == out.shape[-1] == 14
+    # a valid latent index maps to a bbox inside the image
+    box = compute_rf_prototype(224, [0, 13, 13], info)
+    assert 0 <= box[1] <= box[2] <= 224 and 0 <= box[3] <= box[4] <= 224
+
+
+@pytest.mark.parametrize('name', ['Proxy_Anchor', 'Proxy_NCA', 'MS',
+                                  'Contrastive', 'Triplet', 'NPair'])
+def test_aux_losses_forward_backward(name):
+    from mgproto_amd.losses import build_aux_loss
+    torch.manual_seed(0)
+    crit = build_aux_loss(name, nb_classes=5, sz_embed=8)
+    x = torch.randn(12, 8, requires_grad=True)
+    t = torch.tensor([0, 0, 1, 1, 2, 2, 3, 3, 4, 4, 0, 1])
+    loss = crit(x, t)
+    assert torch.isfinite(loss)
+    if loss.requires_grad:
+        loss.backward()
+        assert torch.isfinite(x.grad).all()
+
+
+def test_unknown_aux_loss_raises():
+    from mgproto_amd.losses import build_aux_loss
+    with pytest.raises(ValueError):
+        build_aux_loss('nope', nb_classes=2, sz_embed=4)
