@@ -158,3 +158,22 @@ def test_fused_bn_graph_capturable():
         fn()
     g.replay()
     torch.cuda.synchronize()
+
+
+def test_gemm_conv1x1_parity_optin():
+    """GemmConv2d's hipBLASLt path (opt-in) vs F.conv2d."""
+    from mgproto_amd.models.conv1x1 import GemmConv2d
+    torch.manual_seed(0)
+    conv = GemmConv2d(64, 128, kernel_size=1, bias=True).cuda().to(torch.bfloat16)
+    x = torch.randn(4, 64, 14, 14).cuda().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    want = F.conv2d(x, conv.weight, conv.bias)
+    os.environ['MGPROTO_GEMM_CONV1X1'] = '1'
+    try:
+        x1 = x.clone().requires_grad_(True)
+        got = conv(x1)
+        assert torch.allclose(got.float(), want.float(), atol=3e-2, rtol=3e-2)
+        got.float().sum().backward()
+        assert torch.isfinite(conv.weight.grad.float()).all()
+    finally:
+        os.environ.pop('MGPROTO_GEMM_CONV1X1')

@@ -185,10 +185,30 @@ def main():
         start_epoch = state['epoch'] + 1
         log(f'resumed from {args.resume} at epoch {start_epoch}')
 
+    # failure handling: checkpoint on SIGTERM/SIGUSR1 (preemption-safe;
+    # the reference loses everything past the last conditional save)
+    import signal
+
+    def _save_now(signum, frame):
+        if rank == 0:
+            save_train_state(os.path.join(model_dir, 'preempt.pth'), ppnet,
+                             {'joint': joint_optimizer, 'warm': warm_optimizer},
+                             {'joint_lr': joint_lr_scheduler},
+                             epoch=_save_now.epoch,
+                             extra={'reason': f'signal {signum}'})
+            log(f'checkpointed on signal {signum}')
+    _save_now.epoch = start_epoch - 1
+    for sig in (signal.SIGTERM, signal.SIGUSR1):
+        try:
+            signal.signal(sig, _save_now)
+        except (ValueError, OSError):
+            pass
+
     log('start training')
     decay_epochs = cfg.lr_decay_epochs()
     epoch = start_epoch
     for epoch in range(start_epoch, cfg.num_train_epochs):
+        _save_now.epoch = epoch
         log('epoch: \t{0}'.format(epoch))
         use_mining = epoch >= cfg.mine_start
         update_GMM = (epoch >= cfg.updateGMM_start
