@@ -63,28 +63,31 @@ class _GMMScore(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, feat, W, bias, apply_exp):
+    def forward(ctx, feat, Wt, bias, apply_exp):
+        # Wt is [P, 2d] (the fwd kernel's B-panel layout); the bwd kernel
+        # streams the [2d, P] transpose, materialized once per call (~1 MB)
         ext = _native_or_die() if feat.is_cuda else None
         if ext is not None:
-            out = ext.gmm_fwd(feat, W, bias, apply_exp)
+            out = ext.gmm_fwd(feat, Wt, bias, apply_exp)
         else:
             out = torch.addmm(bias.unsqueeze(0),
-                              torch.cat([feat, feat * feat], dim=1), W)
+                              torch.cat([feat, feat * feat], dim=1), Wt.t())
             if apply_exp:
                 out = torch.exp(out)
-        ctx.save_for_backward(feat, W, out if apply_exp else torch.empty(0))
+        ctx.save_for_backward(feat, Wt, out if apply_exp else torch.empty(0))
         ctx.apply_exp = apply_exp
         return out
 
     @staticmethod
     def backward(ctx, grad_out):
-        feat, W, probs = ctx.saved_tensors
+        feat, Wt, probs = ctx.saved_tensors
         g = grad_out * probs if ctx.apply_exp else grad_out
         ext = _native_or_die() if feat.is_cuda else None
         if ext is not None:
-            grad_feat = ext.gmm_bwd(g.contiguous(), feat, W)
+            grad_feat = ext.gmm_bwd(g.contiguous(), feat,
+                                    Wt.t().contiguous())
         else:
-            gw = g @ W.t()                       # [N, 2d]
+            gw = g @ Wt                          # [N, 2d]
             d = feat.shape[1]
             grad_feat = gw[:, :d] + 2.0 * feat * gw[:, d:]
         return grad_feat, None, None, None
@@ -100,8 +103,8 @@ def gmm_scores(feat: torch.Tensor, means: torch.Tensor, covs: torch.Tensor,
     """
     means2 = means.reshape(-1, means.shape[-1]).detach()
     covs2 = covs.reshape(-1, covs.shape[-1]).detach()
-    W, bias = gmm_expand_params(means2.float(), covs2.float(), eps)
-    return _GMMScore.apply(feat, W, bias, apply_exp)
+    Wt, bias = gmm_expand_params(means2.float(), covs2.float(), eps)
+    return _GMMScore.apply(feat, Wt, bias, apply_exp)
 
 
 class _TopkHW(torch.autograd.Function):

@@ -81,25 +81,20 @@ void gmm_fwd_kernel(const float* __restrict__ x,   // [N, d]
             dst2[2] = v.z * v.z; dst2[3] = v.w * v.w;
         }
     }
-    // ---- stage B transposed: Bs[p - p0][j2] = w[j2][p] ------------------
+    // ---- stage B: Bs[p - p0][j2] = wt[p][j2] (wt is [P, 2d]) ------------
+    // row-major copy: coalesced float4 global reads, sequential LDS writes
+    // (staging from the [2d, P] layout needed a transposed write pattern
+    // that measured 371M LDS bank-conflict cycles per bench)
     {
-        for (int t = tid; t < K2 * (BN / 4); t += 256) {
-            const int j2 = t / (BN / 4);
-            const int c4 = t % (BN / 4);
-            const int p = p0 + c4 * 4;
-            float4 v;
-            if (p + 3 < P) {
-                v = *reinterpret_cast<const float4*>(w + (long)j2 * P + p);
-            } else {
-                v.x = (p + 0 < P) ? w[(long)j2 * P + p + 0] : 0.f;
-                v.y = (p + 1 < P) ? w[(long)j2 * P + p + 1] : 0.f;
-                v.z = (p + 2 < P) ? w[(long)j2 * P + p + 2] : 0.f;
-                v.w = (p + 3 < P) ? w[(long)j2 * P + p + 3] : 0.f;
-            }
-            Bs[(c4 * 4 + 0) * KS + j2] = v.x;
-            Bs[(c4 * 4 + 1) * KS + j2] = v.y;
-            Bs[(c4 * 4 + 2) * KS + j2] = v.z;
-            Bs[(c4 * 4 + 3) * KS + j2] = v.w;
+        const int vec_per_row = K2 / 4;
+        for (int t = tid; t < BN * vec_per_row; t += 256) {
+            const int pr = t / vec_per_row;
+            const int c4 = t % vec_per_row;
+            const int p = min(p0 + pr, P - 1);
+            const float4 v = reinterpret_cast<const float4*>(
+                w + (long)p * K2)[c4];
+            float* dst = Bs + pr * KS + c4 * 4;
+            dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
         }
     }
     __syncthreads();
@@ -425,26 +420,25 @@ void argmax_hw_kernel(const float* __restrict__ probs,
 
 static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
 
-torch::Tensor gmm_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias,
+torch::Tensor gmm_fwd(torch::Tensor x, torch::Tensor wt, torch::Tensor bias,
                       bool apply_exp) {
-    CHECK_IN(x); CHECK_IN(w); CHECK_IN(bias);
+    CHECK_IN(x); CHECK_IN(wt); CHECK_IN(bias);
     TORCH_CHECK(x.dtype() == torch::kFloat32, "gmm_fwd: fp32 only");
-    const int N = x.size(0), d = x.size(1), P = w.size(1);
-    TORCH_CHECK(w.size(0) == 2 * d, "w must be [2d, P]");
+    const int N = x.size(0), d = x.size(1), P = wt.size(0);
+    TORCH_CHECK(wt.size(1) == 2 * d, "wt must be [P, 2d]");
     TORCH_CHECK(d % 8 == 0 && d <= 128, "d must be multiple of 8, <= 128");
-    TORCH_CHECK(P % 4 == 0, "P must be a multiple of 4");
     auto out = torch::empty({N, P}, x.options());
     auto stream = at::hip::getCurrentHIPStream();
     if (d <= 64) {
         dim3 grid(ceil_div(N, 128), ceil_div(P, 128));
         hipLaunchKernelGGL((gmm_fwd_kernel<128, 128, 128>), grid, dim3(256), 0,
-                           stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                           stream, x.data_ptr<float>(), wt.data_ptr<float>(),
                            bias.data_ptr<float>(), out.data_ptr<float>(),
                            N, d, P, (int)apply_exp);
     } else {
         dim3 grid(ceil_div(N, 64), ceil_div(P, 64));
         hipLaunchKernelGGL((gmm_fwd_kernel<64, 64, 256>), grid, dim3(256), 0,
-                           stream, x.data_ptr<float>(), w.data_ptr<float>(),
+                           stream, x.data_ptr<float>(), wt.data_ptr<float>(),
                            bias.data_ptr<float>(), out.data_ptr<float>(),
                            N, d, P, (int)apply_exp);
     }

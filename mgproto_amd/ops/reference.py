@@ -41,9 +41,10 @@ def gmm_expand_params(means: torch.Tensor, covs: torch.Tensor, eps: float = 0.0
         B[p,j]   = -1 / (2 sigma[p,j]^2)
         bias[p]  = -d/2 log(2 pi) - sum_j log sigma[p,j] - sum_j mu[p,j]^2/(2 sigma[p,j]^2)
 
-    Returns ``W [2d, P]`` (A on top of B, GEMM-ready) and ``bias [P]``. This
-    is the exact expression the HIP MFMA kernel evaluates (one [N,2d]x[2d,P]
-    GEMM over [x, x^2]).
+    Returns ``Wt [P, 2d]`` ([A | B] rows) and ``bias [P]``. This is the
+    exact expression the HIP MFMA kernel evaluates (one [N,2d]x[2d,P] GEMM
+    over [x, x^2]); the row-major [P, 2d] layout lets the kernel stage its
+    B panel with coalesced reads and conflict-free LDS writes.
     """
     P = means.shape[0]
     d = means.shape[1]
@@ -54,8 +55,8 @@ def gmm_expand_params(means: torch.Tensor, covs: torch.Tensor, eps: float = 0.0
     bias = (-0.5 * d * LOG_2PI
             - torch.log(sig).sum(dim=1)
             - 0.5 * (means * means * inv_var).sum(dim=1))   # [P]
-    W = torch.cat([A, Bq], dim=1).t().contiguous()   # [2d, P]
-    return W, bias
+    Wt = torch.cat([A, Bq], dim=1).contiguous()      # [P, 2d]
+    return Wt, bias
 
 
 def gmm_logprob(feat: torch.Tensor, means: torch.Tensor, covs: torch.Tensor,
@@ -71,9 +72,9 @@ def gmm_logprob(feat: torch.Tensor, means: torch.Tensor, covs: torch.Tensor,
     """
     means = means.reshape(-1, means.shape[-1]).detach()
     covs = covs.reshape(-1, covs.shape[-1]).detach()
-    W, bias = gmm_expand_params(means, covs, eps)
+    Wt, bias = gmm_expand_params(means, covs, eps)
     x2 = torch.cat([feat, feat * feat], dim=1)       # [N, 2d]
-    return x2 @ W + bias
+    return x2 @ Wt.t() + bias
 
 
 def gmm_logprob_direct(feat: torch.Tensor, means: torch.Tensor,

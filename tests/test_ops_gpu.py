@@ -36,14 +36,14 @@ def make_gmm(N, P, d, device, seed=0, uniform_sigma=True):
 def test_gmm_fwd_parity(N, P, d):
     dev = torch.device('cuda')
     feat, means, covs = make_gmm(N, P, d, dev)
-    W, bias = R.gmm_expand_params(means, covs)
+    Wt, bias = R.gmm_expand_params(means, covs)
     ext = _ext()
-    out = ext.gmm_fwd(feat, W.contiguous(), bias.contiguous(), True)
+    out = ext.gmm_fwd(feat, Wt.contiguous(), bias.contiguous(), True)
     want = torch.exp(R.gmm_logprob(feat, means, covs))
     assert torch.allclose(out, want, atol=1e-4, rtol=1e-4), \
         (out - want).abs().max().item()
     # no-exp variant
-    out_lp = ext.gmm_fwd(feat, W.contiguous(), bias.contiguous(), False)
+    out_lp = ext.gmm_fwd(feat, Wt.contiguous(), bias.contiguous(), False)
     want_lp = R.gmm_logprob(feat, means, covs)
     assert torch.allclose(out_lp, want_lp, atol=1e-4, rtol=1e-4)
 
@@ -52,8 +52,8 @@ def test_gmm_fwd_vs_direct_oracle():
     """Against the literal (x-mu)^2/sigma^2 formula, non-uniform sigma."""
     dev = torch.device('cuda')
     feat, means, covs = make_gmm(4096, 2000, 64, dev, uniform_sigma=False)
-    W, bias = R.gmm_expand_params(means, covs)
-    out = _ext().gmm_fwd(feat, W.contiguous(), bias.contiguous(), False)
+    Wt, bias = R.gmm_expand_params(means, covs)
+    out = _ext().gmm_fwd(feat, Wt.contiguous(), bias.contiguous(), False)
     want = R.gmm_logprob_direct(feat, means, covs)
     assert torch.allclose(out, want, atol=2e-4, rtol=1e-4), \
         (out - want).abs().max().item()
@@ -64,10 +64,10 @@ def test_gmm_fwd_vs_direct_oracle():
 def test_gmm_bwd_parity(N, P, d):
     dev = torch.device('cuda')
     feat, means, covs = make_gmm(N, P, d, dev, seed=3)
-    W, bias = R.gmm_expand_params(means, covs)
+    Wt, bias = R.gmm_expand_params(means, covs)
     g = torch.randn(N, P, device=dev)
-    got = _ext().gmm_bwd(g.contiguous(), feat, W.contiguous())
-    gw = g @ W.t()
+    got = _ext().gmm_bwd(g.contiguous(), feat, Wt.t().contiguous())
+    gw = g @ Wt
     want = gw[:, :d] + 2.0 * feat * gw[:, d:]
     assert torch.allclose(got, want, atol=1e-3, rtol=1e-4), \
         (got - want).abs().max().item()
