@@ -10,7 +10,6 @@ reference's ((img, label), (path, label)) format (utils/helpers.py:8).
 import os
 from typing import Callable, List, Optional, Tuple
 
-import torch
 from torch.utils.data import Dataset
 
 IMG_EXTENSIONS = ('.jpg', '.jpeg', '.png', '.ppm', '.bmp', '.pgm', '.tif',

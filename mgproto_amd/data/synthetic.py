@@ -9,8 +9,6 @@ memory for benchmarking — the bench measures the training step, not host
 RNG.
 """
 
-from typing import Optional
-
 import torch
 from torch.utils.data import Dataset
 

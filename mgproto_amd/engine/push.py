@@ -99,6 +99,11 @@ def push_prototypes(dataloader,
         image, label, idx, _path = _batch_fields(item)
         B = image.shape[0]
         if idx is None:
+            if comm is not None and comm.is_distributed:
+                raise ValueError(
+                    'distributed push needs datasets that yield a GLOBAL '
+                    'image index per item ((img, label, idx) format) — '
+                    'rank-local offsets cannot identify images across ranks')
             idx = torch.arange(batch_offset, batch_offset + B)
         batch_offset += B
         search_batch = (preprocess_input_function(image)

@@ -28,7 +28,6 @@ Deliberate behavioural divergences (each rank-invariant by construction):
 """
 
 import math
-from typing import Optional, Tuple
 
 import torch
 import torch.nn as nn

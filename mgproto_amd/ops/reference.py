@@ -18,10 +18,9 @@ this layout.
 """
 
 import math
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
-import torch.nn.functional as F
 
 LOG_2PI = math.log(2.0 * math.pi)
 

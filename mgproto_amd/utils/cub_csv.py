@@ -13,9 +13,7 @@ import os
 from typing import Dict, List
 
 import numpy as np
-import pandas as pd
 import torch
-import torch.nn.functional as F
 
 
 def get_img_coordinates(img_size, softmaxes_shape, patchsize, skip,

@@ -221,6 +221,8 @@ class MemoryBank(nn.Module):
                 self.mem[c].copy_(state_dict[key])
                 found += 1
                 state_dict.pop(key)
+            elif strict:
+                missing_keys.append(key)
         key = prefix + 'mem_len'
         if key in state_dict:
             self.mem_len.copy_(state_dict[key])

@@ -7,7 +7,7 @@ an input-pixel bounding box.
 """
 
 import math
-from typing import List, Sequence, Tuple
+from typing import List, Sequence
 
 
 def compute_layer_rf_info(layer_filter_size, layer_stride, layer_padding,
