@@ -102,8 +102,10 @@ def main():
 
     # whole-step hipGraph capture: single-node-per-rank eager graphs replace
     # the reference's (nonexistent) tracing compiler — cuts ~2k kernel-launch
-    # overheads per step.  Off for multi-rank (RCCL capture handled later).
-    use_graph = (device.type == 'cuda') and not args.no_graph and world == 1
+    # overheads per step.  Multi-rank capture (RCCL collectives inside the
+    # graph) is opt-in via MGPROTO_GRAPH_DIST=1 pending 8-GPU validation.
+    use_graph = (device.type == 'cuda') and not args.no_graph and (
+        world == 1 or os.environ.get('MGPROTO_GRAPH_DIST') == '1')
 
     aux = build_aux_loss('Proxy_Anchor', nb_classes=C, sz_embed=32).to(device)
     groups = [

@@ -87,6 +87,9 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
     mem_nonempty = bool(int(m.queue.mem_len.sum()) > 0) if update_GMM else False
     iter_base = int(m.iteration_counter)
 
+    from ..utils.timing import PhaseTimer
+    timer = PhaseTimer(device=device)   # enabled iff MGPROTO_TIMING=1
+
     for i, batch in enumerate(dataloader):
         image, label = batch[0], batch[1]
         image = image.to(device, non_blocking=True)
@@ -95,18 +98,21 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
         if em_runner is not None:
             em_runner.sync()  # next forward reads EM-updated means/priors
 
-        with _amp_ctx(device, amp_dtype):
+        with timer.phase('forward'), _amp_ctx(device, amp_dtype):
             output, x_auxiliary = model(image, target)
 
-        output = output.float()
-        if use_mine and output.shape[2] > 1:
-            mine_loss = sum(F.cross_entropy(output[:, :, k], target)
-                            for k in range(1, output.shape[2])) / (output.shape[2] - 1)
-        else:
-            mine_loss = torch.zeros((), device=device)
-        cross_entropy = F.cross_entropy(output[:, :, 0], target)
-        aux_loss = (aux_criterion(x_auxiliary.float(), target)
-                    if aux_criterion is not None else torch.zeros((), device=device))
+        with timer.phase('loss'):
+            output = output.float()
+            if use_mine and output.shape[2] > 1:
+                mine_loss = sum(F.cross_entropy(output[:, :, k], target)
+                                for k in range(1, output.shape[2])) \
+                    / (output.shape[2] - 1)
+            else:
+                mine_loss = torch.zeros((), device=device)
+            cross_entropy = F.cross_entropy(output[:, :, 0], target)
+            aux_loss = (aux_criterion(x_auxiliary.float(), target)
+                        if aux_criterion is not None
+                        else torch.zeros((), device=device))
 
         predicted = torch.argmax(output[:, :, 0].detach(), dim=1)
         n_examples += target.numel()
@@ -118,17 +124,27 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
 
         loss = (coefs['crs_ent'] * cross_entropy + coefs['mine'] * mine_loss
                 + coefs['aux'] * aux_loss)
-        optimizer.zero_grad(set_to_none=True)
-        loss.backward()
-        optimizer.step()
+        with timer.phase('backward'):
+            optimizer.zero_grad(set_to_none=True)
+            loss.backward()
+        with timer.phase('optimizer'):
+            optimizer.step()
 
         # EM update (reference train_and_test.py:61-63; update_interval=1)
         if update_GMM:
             mem_nonempty = mem_nonempty or (i == 0 and target.numel() > 0)
             if mem_nonempty and (iter_base + i + 1) % m.update_interval == 0:
-                (em_runner.run() if em_runner is not None else m.update_GMM())
+                with timer.phase('em'):
+                    (em_runner.run() if em_runner is not None
+                     else m.update_GMM())
 
         if print_every and i % print_every == 0:
+            if timer.enabled:
+                phases = timer.summary()
+                print('  phase ms: ' + '  '.join(
+                    f'{k}={v:.2f}' for k, v in phases.items()))
+                if metrics is not None:
+                    metrics.log({f'time/{k}': v for k, v in phases.items()})
             full_ratio = (m.queue.mem_len == m.capacity_pc).float().mean()
             acc = float(n_correct) / (float(n_examples) + 1e-6) * 100
             print(f'{i} {len(dataloader)} \tLoss: {float(loss):.4f} '

@@ -177,3 +177,37 @@ def test_mining_grad_flows_to_backbone(small_model):
     assert conv1_grad is not None and conv1_grad.abs().sum() > 0
     # prototype means get no grad from the CE path (detached, model.py:264)
     assert m.prototype_means.grad is None or m.prototype_means.grad.abs().sum() == 0
+
+
+def test_forward_matches_manual_pipeline():
+    """Full-forward differential oracle: model.forward(x, None) must equal
+    the manually composed pipeline (direct log-prob formula -> exp -> top-T
+    -> pi-weighted mixture -> log), i.e. the reference's semantics
+    (model.py:208-254) assembled from first principles."""
+    from mgproto_amd.ops import reference as R
+    torch.manual_seed(0)
+    C, K, d, T = 5, 3, 16, 4
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(C * K, d, 1, 1), num_classes=C,
+                          add_on_layers_type='regular', sz_embedding=8,
+                          mem_capacity=4, mine_K=T)
+    m.eval()
+    x = torch.randn(2, 3, 64, 64)
+    with torch.no_grad():
+        got, _ = m(x, None)
+
+        # manual pipeline from push_forward pieces + direct formula
+        base, _ = m.conv_features(x)
+        base = torch.nn.functional.normalize(base, dim=1)
+        B, dd, H, W = base.shape
+        feat = base.permute(0, 2, 3, 1).reshape(-1, dd)
+        lp = R.gmm_logprob_direct(feat, m.prototype_means.data,
+                                  m.prototype_covs.data)      # [N, P]
+        probs = lp.exp().view(B, H * W, C * K)
+        vals, _ = torch.topk(probs, T, dim=1)                 # [B, T, P]
+        vals = vals.permute(0, 2, 1).view(B, C, K, T)
+        pi = m.last_layer.weight.data.view(C, C, K)[torch.arange(C),
+                                                    torch.arange(C)]
+        want = torch.log(torch.einsum('bckt,ck->bct', vals, pi))
+    assert torch.allclose(got, want, atol=1e-4, rtol=1e-4), \
+        (got - want).abs().max().item()
