@@ -51,6 +51,8 @@ class BucketedGradReducer:
 
         self._bucket_of = {}
         self._pending = [0] * len(self.buckets)
+        self._ready = [False] * len(self.buckets)
+        self._next_launch = len(self.buckets)
         self._flat: List[Optional[torch.Tensor]] = [None] * len(self.buckets)
         for bi, bucket in enumerate(self.buckets):
             for p in bucket:
@@ -67,6 +69,8 @@ class BucketedGradReducer:
         if not self.enabled:
             return
         self._works.clear()
+        self._ready = [False] * len(self.buckets)
+        self._next_launch = 0
         for bi, bucket in enumerate(self.buckets):
             self._pending[bi] = len(bucket)
 
@@ -74,7 +78,15 @@ class BucketedGradReducer:
         bi = self._bucket_of[id(p)]
         self._pending[bi] -= 1
         if self._pending[bi] == 0:
-            self._launch(bi)
+            # collectives must be issued in the SAME order on every rank;
+            # autograd completion order is same-graph-deterministic but we
+            # do not rely on it: launch strictly in bucket-index order,
+            # holding back buckets that complete early (as DDP does)
+            self._ready[bi] = True
+            while (self._next_launch < len(self.buckets)
+                   and self._ready[self._next_launch]):
+                self._launch(self._next_launch)
+                self._next_launch += 1
 
     def _launch(self, bi):
         bucket = self.buckets[bi]
@@ -93,6 +105,8 @@ class BucketedGradReducer:
         # Partially-filled buckets (params frozen mid-run, or unused in this
         # graph) never auto-launched: reduce their present grads now so no
         # rank diverges. Ranks must agree on which params got grads.
+        # (Complete buckets held back behind a never-completing bucket are
+        # launched right after, still in index order.)
         for bi, bucket in enumerate(self.buckets):
             if 0 < self._pending[bi] < len(bucket):
                 present = [p for p in bucket if p.grad is not None]
@@ -105,6 +119,9 @@ class BucketedGradReducer:
                             flat, [p.grad for p in present])):
                         p.grad.copy_(g)
                 self._pending[bi] = 0
+            elif self._ready[bi] and bi >= self._next_launch:
+                self._launch(bi)
+                self._next_launch = bi + 1
         for work, bi in self._works:
             work.wait()
             bucket = self.buckets[bi]
