@@ -109,3 +109,56 @@ def test_push_accepts_reference_format(tmp_path):
     chosen = push_prototypes(loader, model, log=lambda *a: None,
                              preprocess_input_function=preprocess_input_function)
     assert len(chosen) > 0
+
+
+def test_fused_train_transform():
+    from mgproto_amd.data.preprocess import mean, std
+    img = _img(120, 100, seed=3)
+    tf = T.FusedTrainTransform(64, normalize=T.Normalize(mean, std))
+    import random as _r
+    _r.seed(0)
+    outs = [tf(img) for _ in range(4)]
+    for o in outs:
+        assert o.shape == (3, 64, 64)
+        assert torch.isfinite(o).all()
+    # stochastic: different draws differ
+    assert not torch.allclose(outs[0], outs[1])
+
+
+def test_fused_transform_identity_params_matches_resize():
+    """With all randomness disabled and a square image, the fused
+    homography must reduce to a plain resize."""
+    img = _img(80, 80, seed=5)
+    tf = T.FusedTrainTransform(64, scale=(1.0, 1.0), perspective_p=0.0,
+                               jitter=None, degrees=0, shear=(0, 0),
+                               translate=(0, 0))
+    import random as _r
+    for seed in range(3):
+        _r.seed(seed)
+        out = tf(img)
+        want = T.ToTensor()(T.Resize((64, 64))(img))
+        # im.transform point-samples (no antialias) while PIL resize
+        # filters, so downscales differ slightly; bound the gap
+        assert (out - want).abs().mean() < 0.06, \
+            float((out - want).abs().mean())
+
+
+def test_fast_color_jitter_matches_pil_enhance():
+    from PIL import ImageEnhance
+    img = _img(32, 32, seed=7)
+    import random as _r
+
+    # brightness only
+    fj = T.FastColorJitter(brightness=(1.3, 1.3))
+    _r.seed(0)
+    got = np.asarray(fj(img), dtype=np.float32)
+    want = np.asarray(ImageEnhance.Brightness(img).enhance(1.3),
+                      dtype=np.float32)
+    assert np.abs(got - want).mean() < 1.0
+
+    # saturation only
+    fj = T.FastColorJitter(saturation=(0.5, 0.5))
+    _r.seed(0)
+    got = np.asarray(fj(img), dtype=np.float32)
+    want = np.asarray(ImageEnhance.Color(img).enhance(0.5), dtype=np.float32)
+    assert np.abs(got - want).mean() < 2.0
