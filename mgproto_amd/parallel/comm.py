@@ -103,10 +103,16 @@ class Comm:
             dist.barrier()
 
     def broadcast_module(self, module: torch.nn.Module, src: int = 0):
-        """Make parameters+buffers bit-identical across ranks at startup."""
+        """Make parameters+buffers bit-identical across ranks at startup.
+
+        Iterates named_parameters/named_buffers (the LIVE tensors, including
+        non-persistent buffers) — state_dict would hand us the memory bank's
+        exported logical COPIES, which a broadcast would not write through.
+        """
         if not self.is_distributed:
             return
         with torch.no_grad():
-            for p in module.state_dict().values():
+            for _, p in list(module.named_parameters()) \
+                    + list(module.named_buffers()):
                 if torch.is_tensor(p) and p.numel() > 0:
                     self.broadcast(p.data if hasattr(p, 'data') else p, src=src)

@@ -81,5 +81,9 @@ def setup_miopen_db():
     for f in os.listdir(src):
         tgt = os.path.join(dst, f)
         if not os.path.exists(tgt):
-            shutil.copy2(os.path.join(src, f), tgt)
+            # atomic publish: concurrent ranks must never observe a
+            # half-copied db file
+            tmp = tgt + f'.tmp{os.getpid()}'
+            shutil.copy2(os.path.join(src, f), tmp)
+            os.replace(tmp, tgt)
     os.environ['MIOPEN_USER_DB_PATH'] = dst
