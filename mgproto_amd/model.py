@@ -161,11 +161,16 @@ class MGProto(nn.Module):
             self.add_on_layers = nn.Sequential(*add_on_layers)
         elif add_on_layers_type == 'regular_upsample':
             # the reference's commented-out R50/iNat variant (model.py:138):
-            # 2x bilinear upsample -> 28x28 latent grid at 224 input
+            # 2x bilinear upsample -> 28x28 latent grid at 224 input.
+            # The upsample COMMUTES exactly with the (linear, no-activation)
+            # 1x1 convs, so it runs LAST: the convs see the 14x14 grid (4x
+            # less work) and the upsample moves [B,d,..] instead of
+            # [B,2048,..] tensors (32x less traffic; its backward alone was
+            # 1.7 ms/step the other way — profiles/).
             self.add_on_layers = nn.Sequential(
-                nn.Upsample(scale_factor=2, mode='bilinear', align_corners=False),
                 GemmConv2d(first_add_on_layer_in_channels, d, kernel_size=1),
                 GemmConv2d(d, d, kernel_size=1),
+                nn.Upsample(scale_factor=2, mode='bilinear', align_corners=False),
             )
         else:  # 'regular'
             self.add_on_layers = nn.Sequential(

@@ -211,3 +211,32 @@ def test_forward_matches_manual_pipeline():
         want = torch.log(torch.einsum('bckt,ck->bct', vals, pi))
     assert torch.allclose(got, want, atol=1e-4, rtol=1e-4), \
         (got - want).abs().max().item()
+
+
+def test_upsample_conv_commutation():
+    """conv1x1(upsample(x)) == upsample(conv1x1(x)) exactly (the
+    regular_upsample add-on runs the upsample LAST for 32x less traffic;
+    this pins the mathematical equivalence)."""
+    torch.manual_seed(0)
+    Ci, d = 32, 8
+    c1 = torch.nn.Conv2d(Ci, d, 1)
+    c2 = torch.nn.Conv2d(d, d, 1)
+    up = torch.nn.Upsample(scale_factor=2, mode='bilinear', align_corners=False)
+    x = torch.randn(2, Ci, 7, 7)
+    with torch.no_grad():
+        ref = c2(c1(up(x)))          # the reference's commented-out order
+        ours = up(c2(c1(x)))
+    assert torch.allclose(ref, ours, atol=1e-5), \
+        (ref - ours).abs().max().item()
+
+
+def test_regular_upsample_grid():
+    torch.manual_seed(0)
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(10, 16, 1, 1), num_classes=5,
+                          add_on_layers_type='regular_upsample',
+                          sz_embedding=8, mem_capacity=4, mine_K=2)
+    with torch.no_grad():
+        bf, dist = m.push_forward(torch.randn(1, 3, 64, 64))
+    # 64/16 = 4 latent, x2 upsample -> 8x8 grid
+    assert dist.shape[-2:] == (8, 8)
