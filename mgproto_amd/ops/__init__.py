@@ -84,6 +84,14 @@ class _GMMScore(torch.autograd.Function):
         g = grad_out * probs if ctx.apply_exp else grad_out
         ext = _native_or_die() if feat.is_cuda else None
         if ext is not None:
+            P = Wt.shape[0]
+            if P % 4:
+                # the bwd kernel streams g rows with float4 loads; pad the
+                # K dimension (zero columns contribute nothing). Reachable
+                # with e.g. C=37 x K=10 (Pets config).
+                pad = 4 - P % 4
+                g = torch.nn.functional.pad(g, (0, pad))
+                Wt = torch.nn.functional.pad(Wt, (0, 0, 0, pad))
             grad_feat = ext.gmm_bwd(g.contiguous(), feat,
                                     Wt.t().contiguous())
         else:

@@ -32,7 +32,8 @@ def make_gmm(N, P, d, device, seed=0, uniform_sigma=True):
 
 @pytest.mark.parametrize('N,P,d', [(15680, 2000, 64), (1024, 2000, 64),
                                    (62720, 2000, 64), (640, 2000, 128),
-                                   (1000, 500, 64), (128, 96, 64)])
+                                   (1000, 500, 64), (128, 96, 64),
+                                   (512, 370, 64)])  # P%4!=0 (Pets C=37)
 def test_gmm_fwd_parity(N, P, d):
     dev = torch.device('cuda')
     feat, means, covs = make_gmm(N, P, d, dev)
@@ -73,11 +74,12 @@ def test_gmm_bwd_parity(N, P, d):
         (got - want).abs().max().item()
 
 
-def test_gmm_autograd_end_to_end():
+@pytest.mark.parametrize('P', [2000, 370])  # 370: exercises bwd padding
+def test_gmm_autograd_end_to_end(P):
     """Through ops.gmm_scores (native path) vs the CPU reference autograd."""
     from mgproto_amd import ops
     dev = torch.device('cuda')
-    feat, means, covs = make_gmm(1024, 2000, 64, dev, seed=5)
+    feat, means, covs = make_gmm(1024, P, 64, dev, seed=5)
     f_gpu = feat.clone().requires_grad_(True)
     out = ops.gmm_scores(f_gpu, means, covs, apply_exp=True)
     gout = torch.randn_like(out)
