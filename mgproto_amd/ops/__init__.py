@@ -66,7 +66,11 @@ class _GMMScore(torch.autograd.Function):
     def forward(ctx, feat, Wt, bias, apply_exp):
         # Wt is [P, 2d] (the fwd kernel's B-panel layout); the bwd kernel
         # streams the [2d, P] transpose, materialized once per call (~1 MB)
+        d = feat.shape[1]
         ext = _native_or_die() if feat.is_cuda else None
+        if ext is not None and (d % 8 != 0 or d > 128):
+            ext = None   # out of the kernel's envelope: hipBLASLt GEMM path
+        ctx.used_ext = ext is not None
         if ext is not None:
             out = ext.gmm_fwd(feat, Wt, bias, apply_exp)
         else:
@@ -82,7 +86,7 @@ class _GMMScore(torch.autograd.Function):
     def backward(ctx, grad_out):
         feat, Wt, probs = ctx.saved_tensors
         g = grad_out * probs if ctx.apply_exp else grad_out
-        ext = _native_or_die() if feat.is_cuda else None
+        ext = _native_or_die() if (feat.is_cuda and ctx.used_ext) else None
         if ext is not None:
             P = Wt.shape[0]
             if P % 4:
@@ -121,6 +125,8 @@ class _TopkHW(torch.autograd.Function):
     @staticmethod
     def forward(ctx, probs, T):
         ext = _native_or_die() if probs.is_cuda else None
+        if ext is not None and T > 32:
+            ext = None   # kernel register-list limit; torch.topk on GPU
         if ext is not None:
             vals, idx = ext.topk_hw(probs, T)
         else:
