@@ -260,3 +260,7 @@ def impl_metric_allreduce(rank, world):
     v = comm.all_gather_varlen(torch.arange(rank + 1).float())
     assert v.tolist() == [0.0, 0.0, 1.0]
     return 'ok'
+
+
+def test_distributed_push_three_ranks():
+    _run_workers(impl_distributed_push_matches_single, world=3)
