@@ -265,7 +265,7 @@ def _testing_with_OoD(model, dataloaders, class_specific=True, log=print,
 
     results = {'acc': acc}
     for li, loader in enumerate(ood_loaders, start=1):
-        preds = []
+        preds, scores = [], []
         for batch in loader:
             image = batch[0].to(device, non_blocking=True)
             with _amp_ctx(device, amp_dtype):
@@ -273,15 +273,40 @@ def _testing_with_OoD(model, dataloaders, class_specific=True, log=print,
             output_prob = output[:, :, 0].float().exp()
             # reference :213 thresholds the class-MEAN density
             preds.append(output_prob.mean(dim=1) > ood_thresh.to(device))
+            scores.append(output_prob.sum(dim=1))
         preds = torch.cat(preds) if preds else torch.zeros(0, device=device)
+        scores = torch.cat(scores) if scores else torch.zeros(0, device=device)
         if comm is not None:
             preds = comm.all_gather_varlen(preds.float())
+            scores = comm.all_gather_varlen(scores)
         fpr95 = float(preds.float().sum()) / max(preds.numel(), 1)
         log('\tFPR95_{0}: \t{1}'.format(li, fpr95))
         results[f'FPR95_{li}'] = fpr95
+        # AUROC of p(x) as the ID-vs-OoD score (the paper's headline OoD
+        # metric; the reference repo only computes the threshold FPR)
+        auroc = _density_auroc(id_probs.float().cpu(), scores.float().cpu())
+        if auroc is not None:
+            log('\tAUROC_{0}: \t{1}'.format(li, auroc))
+            results[f'AUROC_{li}'] = auroc
         if metrics is not None:
-            metrics.log({f'ood/FPR95_{li}': fpr95})
+            metrics.log({f'ood/FPR95_{li}': fpr95,
+                         **({f'ood/AUROC_{li}': auroc} if auroc is not None
+                            else {})})
     return n_correct / max(n_examples, 1), results
+
+
+def _density_auroc(id_scores, ood_scores):
+    """AUROC with the mixture density p(x) as the in-distribution score."""
+    if id_scores.numel() == 0 or ood_scores.numel() == 0:
+        return None
+    try:
+        from sklearn.metrics import roc_auc_score
+        y = torch.cat([torch.ones_like(id_scores),
+                       torch.zeros_like(ood_scores)]).numpy()
+        s = torch.cat([id_scores, ood_scores]).numpy()
+        return float(roc_auc_score(y, s))
+    except Exception:  # noqa: BLE001  (sklearn absent or degenerate input)
+        return None
 
 
 def train(model, dataloader, optimizer, aux_criterion=None, use_mine=False,

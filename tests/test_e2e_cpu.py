@@ -113,3 +113,13 @@ def test_push_deterministic_reforward():
         want = feats[0, :, h, w]
         got = model.prototype_means.data.view(model.num_prototypes, -1)[j]
         assert torch.allclose(got, want, atol=1e-6)
+
+
+def test_density_auroc():
+    from mgproto_amd.engine.trainer import _density_auroc
+    ident = torch.tensor([5.0, 4.0, 6.0, 5.5])
+    ood = torch.tensor([1.0, 0.5, 2.0])
+    assert _density_auroc(ident, ood) == 1.0        # perfectly separated
+    assert _density_auroc(ident, ident) == 0.5 or \
+        abs(_density_auroc(ident, ident) - 0.5) < 0.2
+    assert _density_auroc(torch.zeros(0), ood) is None

@@ -87,6 +87,9 @@ def main():
     parser.add_argument('--out', type=str, default=None)
     parser.add_argument('--ood-eval', action='store_true')
     parser.add_argument('--addon', type=str, default=None)
+    parser.add_argument('--seed', type=int, default=None,
+                        help='deterministic init/data seeding (the reference '
+                             'ships seeding commented out, main.py:45-49)')
     args = parser.parse_args()
 
     cfg = Settings(base_architecture=args.arch, aux_loss=args.aux_loss,
@@ -110,6 +113,13 @@ def main():
     if args.epochs is not None:
         cfg.num_train_epochs = args.epochs
         cfg.push_epochs = [i for i in range(args.epochs) if i % 10 == 0]
+
+    if args.seed is not None:
+        import random as _random
+        _random.seed(args.seed)
+        torch.manual_seed(args.seed)
+        if torch.cuda.is_available():
+            torch.cuda.manual_seed_all(args.seed)
 
     comm = Comm() if int(os.environ.get('WORLD_SIZE', '1')) > 1 else None
     rank = comm.rank if comm else 0
