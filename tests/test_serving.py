@@ -1,0 +1,80 @@
+"""Serving path: InferenceEngine semantics + the FastAPI surface."""
+
+import io
+
+import numpy as np
+import pytest
+import torch
+import torch.nn.functional as F
+
+from mgproto_amd.model import construct_MGProto
+from mgproto_amd.serving import InferenceEngine, create_app
+
+
+@pytest.fixture(scope='module')
+def engine():
+    torch.manual_seed(0)
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(20, 16, 1, 1), num_classes=5,
+                          add_on_layers_type='regular', sz_embedding=8,
+                          mem_capacity=4, mine_K=3)
+    return InferenceEngine(m, torch.device('cpu'), amp=False)
+
+
+def test_predict_schema(engine):
+    x = torch.randn(2, 3, 64, 64)
+    res = engine.predict(x, topk_classes=3, explain_topk=2)
+    assert len(res) == 2
+    for r in res:
+        assert 0 <= r['pred_class'] < 5
+        assert len(r['top_classes']) == 3
+        probs = [t['prob'] for t in r['top_classes']]
+        assert probs == sorted(probs, reverse=True)
+        assert r['density_pX'] > 0
+        assert len(r['explanations']) == 2
+        for e in r['explanations']:
+            # explanations come from the predicted class's prototypes
+            assert e['prototype'] // 4 == r['pred_class']
+            assert 'rf_bbox_yxyx' in e and len(e['rf_bbox_yxyx']) == 4
+            y1, y2, x1, x2 = e['rf_bbox_yxyx']
+            assert 0 <= y1 <= y2 <= 64 and 0 <= x1 <= x2 <= 64
+
+
+def test_predict_consistent_with_forward(engine):
+    """Serving logits must equal model.forward's level-0 output (eval)."""
+    torch.manual_seed(1)
+    x = torch.randn(2, 3, 64, 64)
+    res = engine.predict(x, topk_classes=5)
+    with torch.no_grad():
+        logits, _ = engine.model(x, None)
+    want = F.softmax(logits[:, :, 0], dim=1)
+    for b in range(2):
+        got = sorted(res[b]['top_classes'], key=lambda t: t['class'])
+        for t in got:
+            assert abs(t['prob'] - float(want[b, t['class']])) < 1e-4
+
+
+def test_fastapi_endpoints(engine):
+    pytest.importorskip('fastapi')
+    from fastapi.testclient import TestClient
+    from PIL import Image
+
+    app = create_app(engine, class_names=[f'bird_{i}' for i in range(5)])
+    client = TestClient(app)
+
+    r = client.get('/healthz')
+    assert r.status_code == 200 and r.json()['status'] == 'ok'
+
+    r = client.get('/model_info')
+    assert r.json()['num_classes'] == 5
+
+    rng = np.random.RandomState(0)
+    buf = io.BytesIO()
+    Image.fromarray(rng.randint(0, 255, (80, 90, 3), dtype=np.uint8)) \
+        .save(buf, format='PNG')
+    r = client.post('/predict', content=buf.getvalue(),
+                    headers={'content-type': 'image/png'})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert 'pred_class' in body and 'explanations' in body
+    assert body['top_classes'][0]['name'].startswith('bird_')
