@@ -78,3 +78,22 @@ def test_fastapi_endpoints(engine):
     body = r.json()
     assert 'pred_class' in body and 'explanations' in body
     assert body['top_classes'][0]['name'].startswith('bird_')
+
+
+@pytest.mark.gpu
+def test_serving_graph_capture_gpu():
+    """Graph-captured serving forward == eager forward on GPU."""
+    torch.manual_seed(0)
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(20, 16, 1, 1), num_classes=5,
+                          add_on_layers_type='regular', sz_embedding=8,
+                          mem_capacity=4, mine_K=3).cuda()
+    m.features = m.features.to(memory_format=torch.channels_last)
+    eng = InferenceEngine(m)
+    x = torch.randn(4, 3, 64, 64)
+    want = eng.predict(x, topk_classes=3)
+    eng.capture(4, 64)
+    got = eng.predict(x, topk_classes=3)
+    for w, g in zip(want, got):
+        assert w['pred_class'] == g['pred_class']
+        assert abs(w['density_pX'] - g['density_pX']) < 1e-3 * (1 + abs(w['density_pX']))

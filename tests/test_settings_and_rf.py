@@ -90,3 +90,83 @@ def test_unknown_aux_loss_raises():
     from mgproto_amd.losses import build_aux_loss
     with pytest.raises(ValueError):
         build_aux_loss('nope', nb_classes=2, sz_embed=4)
+
+
+def test_warm_joint_freezing():
+    from mgproto_amd.engine import warm_only, joint
+    from mgproto_amd.model import construct_MGProto
+    import torch as _t
+    _t.manual_seed(0)
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(10, 16, 1, 1), num_classes=5,
+                          add_on_layers_type='regular', sz_embedding=8,
+                          mem_capacity=4, mine_K=2)
+    warm_only(m, log=lambda *a: None)
+    assert all(not p.requires_grad for p in m.features.parameters())
+    assert all(p.requires_grad for p in m.add_on_layers.parameters())
+    joint(m, log=lambda *a: None)
+    assert all(p.requires_grad for p in m.features.parameters())
+
+
+def test_loggers(tmp_path):
+    from mgproto_amd.utils import create_logger, MetricsLogger
+    log, close = create_logger(str(tmp_path / 'x.log'))
+    for i in range(12):
+        log(f'line {i}')
+    close()
+    assert len(open(tmp_path / 'x.log').readlines()) == 12
+
+    ml = MetricsLogger(str(tmp_path / 'm.jsonl'), rank=0)
+    ml.log({'a': 1.0, 'b': 'txt'})
+    ml.log({'a': 2.0}, step=7)
+    ml.close()
+    import json
+    recs = [json.loads(l) for l in open(tmp_path / 'm.jsonl')]
+    assert recs[0]['a'] == 1.0 and recs[0]['_step'] == 0
+    assert recs[1]['_step'] == 7
+    # rank>0 writes nothing
+    ml2 = MetricsLogger(str(tmp_path / 'n.jsonl'), rank=1)
+    ml2.log({'a': 1})
+    ml2.close()
+    import os as _os
+    assert not _os.path.exists(tmp_path / 'n.jsonl')
+
+
+def test_save_model_w_condition(tmp_path):
+    from mgproto_amd.utils.checkpoint import save_model_w_condition
+    import torch as _t
+
+    class Tiny(_t.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.w = _t.nn.Parameter(_t.zeros(2))
+
+    m = Tiny()
+    save_model_w_condition(m, str(tmp_path), '10nopush', accu=0.8224,
+                           target_accu=0.6, log=lambda *a: None)
+    assert (tmp_path / '10nopush0.8224.pth').is_file()   # reference name fmt
+    save_model_w_condition(m, str(tmp_path), '11nopush', accu=0.5,
+                           target_accu=0.6, log=lambda *a: None)
+    assert not (tmp_path / '11nopush0.5000.pth').exists()
+
+
+def test_training_without_aux_criterion():
+    import torch as _t
+    from torch.utils.data import DataLoader
+    from mgproto_amd.engine import trainer as tnt
+    from mgproto_amd.model import construct_MGProto
+    from mgproto_amd.data import SyntheticImages
+    _t.manual_seed(0)
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(10, 16, 1, 1), num_classes=5,
+                          add_on_layers_type='regular', sz_embedding=8,
+                          mem_capacity=4, mine_K=2)
+    opt = _t.optim.Adam(m.parameters(), lr=1e-4)
+    ds = SyntheticImages(8, 5, 64)
+    loader = DataLoader(ds, batch_size=4,
+                        collate_fn=lambda b: (_t.stack([x[0] for x in b]),
+                                              _t.tensor([x[1] for x in b])))
+    acc, res = tnt.train(m, loader, opt, aux_criterion=None, use_mine=True,
+                         coefs={'crs_ent': 1, 'mine': 0.2, 'aux': 0.5},
+                         log=lambda *a: None, amp_dtype='off', print_every=0)
+    assert res['aux_loss'] == 0.0
