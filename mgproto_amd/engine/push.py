@@ -174,7 +174,6 @@ def push_prototypes(dataloader,
             batch = preprocess_input_function(batch)
         batch = batch.to(device, non_blocking=True)
         feats, dist = model.push_forward(batch)          # [b, d, H, W]
-        W_lat = feats.shape[3]
         for bi, (j, img_idx, h, w) in enumerate(blk):
             updates[j] = feats[bi, :, h, w]
             have[j] = 1.0
