@@ -42,25 +42,35 @@ def _usable(x: torch.Tensor, residual) -> bool:
     return _ext() is not None
 
 
+def _use_mask() -> bool:
+    """EXPERIMENTAL (round 2, MGPROTO_BN_MASK=1): the forward emits a 1-bit
+    relu mask per element so the backward never re-reads y (saves ~2 bf16
+    activation passes of backward traffic). Bit-identical semantics to the
+    y>0 test; pending GPU validation."""
+    return os.environ.get('MGPROTO_BN_MASK') == '1'
+
+
 class _FusedBN(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x2, weight, bias, running_mean, running_var,
                 training, momentum, eps, relu, residual2):
-        y, mean, rstd = _ext().bn_fwd(
+        use_mask = relu and _use_mask()
+        y, mean, rstd, mask = _ext().bn_fwd(
             x2, weight.float().contiguous(), bias.float().contiguous(),
             running_mean, running_var, training, momentum, eps, relu,
-            residual2)
-        ctx.save_for_backward(x2, y, weight, mean, rstd)
-        ctx.flags = (training, relu, residual2 is not None)
+            residual2, use_mask)
+        # with the mask, y need not be saved at all (less activation memory)
+        ctx.save_for_backward(x2, mask if use_mask else y, weight, mean, rstd)
+        ctx.flags = (training, relu, residual2 is not None, use_mask)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x2, y, weight, mean, rstd = ctx.saved_tensors
-        training, relu, has_res = ctx.flags
+        x2, y_or_mask, weight, mean, rstd = ctx.saved_tensors
+        training, relu, has_res, use_mask = ctx.flags
         dx, dw, db, dres = _ext().bn_bwd(
-            dy.contiguous(), y, x2, weight.float().contiguous(), mean, rstd,
-            training, relu, has_res)
+            dy.contiguous(), y_or_mask, x2, weight.float().contiguous(),
+            mean, rstd, training, relu, has_res, use_mask)
         return (dx, dw.to(weight.dtype), db.to(weight.dtype), None, None,
                 None, None, None, None, dres if has_res else None)
 

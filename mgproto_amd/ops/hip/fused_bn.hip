@@ -198,12 +198,16 @@ void bn_eval_coeffs_kernel(const float* __restrict__ weight,
 // apply: y = [relu](x*scale + shift [+ res]); RELU/RES are compile-time
 // channel-resident thread mapping (as bn_stats): fixed c8 per thread, row
 // loop — no per-element integer division, coalesced 16B lanes.
-template <bool RELU, bool RES>
+// MASK (EXPERIMENTAL, round-2: MGPROTO_BN_MASK=1): additionally emit one
+// relu-mask byte per 8-channel chunk so the backward never re-reads y
+// (saves 2 bf16 activation passes of backward traffic).
+template <bool RELU, bool RES, bool MASK = false>
 __global__ __launch_bounds__(256)
 void bn_apply_kernel(const short* __restrict__ x,
                      const short* __restrict__ res,
                      const float* __restrict__ scale_shift,
-                     short* __restrict__ y, long M, int C) {
+                     short* __restrict__ y,
+                     unsigned char* __restrict__ mask, long M, int C) {
     const int tpr = C / 8;
     int c8, rsub, rpi;
     if (tpr >= 256) {
@@ -228,14 +232,19 @@ void bn_apply_kernel(const short* __restrict__ x,
         short8 r;
         if (RES) r = *reinterpret_cast<const short8*>(res + m * C + c8);
         short8 o;
+        unsigned char mb = 0;
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
             float f = fmaf(bf2f(v[i]), sc[i], sh[i]);
             if (RES) f += bf2f(r[i]);
             if (RELU) f = fmaxf(f, 0.f);
             o[i] = f2bf(f);
+            // mask from the ROUNDED value: bit-identical to the y>0 test
+            // the non-mask backward performs on the stored bf16 y
+            if (MASK && RELU && bf2f(o[i]) > 0.f) mb |= (1u << i);
         }
         *reinterpret_cast<short8*>(y + m * C + c8) = o;
+        if (MASK) mask[m * (C / 8) + (c8 >> 3)] = mb;
     }
 }
 
@@ -245,15 +254,17 @@ void bn_apply_kernel(const short* __restrict__ x,
 
 // reduce: sum(dy_eff), sum(dy_eff * xhat) per channel; dy_eff = dy * (y>0)
 // when the forward fused a ReLU
-template <bool RELU>
+template <bool RELU, bool MASK = false>
 __global__ __launch_bounds__(256)
 void bn_bwd_reduce_kernel(const short* __restrict__ dy,
-                          const short* __restrict__ y,
+                          const short* __restrict__ y,   // or mask if MASK
                           const short* __restrict__ x,
                           const float* __restrict__ save_mean,
                           const float* __restrict__ save_rstd,
                           float* __restrict__ partials,  // [nblocks, 2C]
                           long M, int C) {
+    const unsigned char* __restrict__ msk =
+        reinterpret_cast<const unsigned char*>(y);
     __shared__ __attribute__((aligned(16))) float lds[256 * 17];
     const int tpr = C / 8;
     float sd[8] = {0}, sx[8] = {0};
@@ -272,11 +283,14 @@ void bn_bwd_reduce_kernel(const short* __restrict__ dy,
                 const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
                 const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
                 short8 yv;
-                if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+                unsigned char mb = 0xff;
+                if (RELU && MASK) mb = msk[m * (C / 8) + (c8 >> 3)];
+                else if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
                 #pragma unroll
                 for (int i = 0; i < 8; ++i) {
                     float gf = bf2f(g[i]);
-                    if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
+                    if (RELU && MASK) { if (!((mb >> i) & 1)) gf = 0.f; }
+                    else if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
                     const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
                     sd[i] += gf; sx[i] += gf * xhat;
                 }
@@ -304,11 +318,14 @@ void bn_bwd_reduce_kernel(const short* __restrict__ dy,
             const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
             const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
             short8 yv;
-            if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+            unsigned char mb = 0xff;
+            if (RELU && MASK) mb = msk[m * (C / 8) + (c8 >> 3)];
+            else if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
             #pragma unroll
             for (int i = 0; i < 8; ++i) {
                 float gf = bf2f(g[i]);
-                if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
+                if (RELU && MASK) { if (!((mb >> i) & 1)) gf = 0.f; }
+                else if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
                 const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
                 sd[i] += gf; sx[i] += gf * xhat;
             }
@@ -319,10 +336,10 @@ void bn_bwd_reduce_kernel(const short* __restrict__ dy,
 
 // apply: dx = gamma*rstd * (dy_eff - sum_dy/M - xhat*sum_dyxhat/M)
 //        d_res = dy_eff (residual branch grad) when RES
-template <bool RELU, bool RES>
+template <bool RELU, bool RES, bool MASK = false>
 __global__ __launch_bounds__(256)
 void bn_bwd_apply_kernel(const short* __restrict__ dy,
-                         const short* __restrict__ y,
+                         const short* __restrict__ y,   // or mask if MASK
                          const short* __restrict__ x,
                          const float* __restrict__ save_mean,
                          const float* __restrict__ save_rstd,
@@ -331,6 +348,8 @@ void bn_bwd_apply_kernel(const short* __restrict__ dy,
                          short* __restrict__ dx,
                          short* __restrict__ dres,
                          long M, int C) {
+    const unsigned char* __restrict__ msk =
+        reinterpret_cast<const unsigned char*>(y);
     const float invM = 1.f / (float)M;
     const int tpr = C / 8;
     int c8, rsub, rpi;
@@ -358,12 +377,15 @@ void bn_bwd_apply_kernel(const short* __restrict__ dy,
         const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
         const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
         short8 yv;
-        if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+        unsigned char mb = 0xff;
+        if (RELU && MASK) mb = msk[m * (C / 8) + (c8 >> 3)];
+        else if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
         short8 odx, odr;
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
             float gf = bf2f(g[i]);
-            if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
+            if (RELU && MASK) { if (!((mb >> i) & 1)) gf = 0.f; }
+            else if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
             if (RES) odr[i] = f2bf(gf);
             const float xhat = (bf2f(xv[i]) - mean[i]) * rstd[i];
             odx[i] = f2bf(gw[i] * (gf - sdy[i] - xhat * sxh[i]));
@@ -375,15 +397,17 @@ void bn_bwd_apply_kernel(const short* __restrict__ dy,
 
 // eval-mode backward (no batch-stat dependency):
 // dx = gamma*rstd*dy_eff
-template <bool RELU, bool RES>
+template <bool RELU, bool RES, bool MASK = false>
 __global__ __launch_bounds__(256)
 void bn_bwd_eval_kernel(const short* __restrict__ dy,
-                        const short* __restrict__ y,
+                        const short* __restrict__ y,   // or mask if MASK
                         const float* __restrict__ save_rstd,
                         const float* __restrict__ weight,
                         short* __restrict__ dx,
                         short* __restrict__ dres,
                         long M, int C) {
+    const unsigned char* __restrict__ msk =
+        reinterpret_cast<const unsigned char*>(y);
     const int tpr = C / 8;
     int c8, rsub, rpi;
     if (tpr >= 256) {
@@ -404,12 +428,15 @@ void bn_bwd_eval_kernel(const short* __restrict__ dy,
          m += (long)gridDim.x * rpi) {
         const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
         short8 yv;
-        if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
+        unsigned char mb = 0xff;
+        if (RELU && MASK) mb = msk[m * (C / 8) + (c8 >> 3)];
+        else if (RELU) yv = *reinterpret_cast<const short8*>(y + m * C + c8);
         short8 odx, odr;
         #pragma unroll
         for (int i = 0; i < 8; ++i) {
             float gf = bf2f(g[i]);
-            if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
+            if (RELU && MASK) { if (!((mb >> i) & 1)) gf = 0.f; }
+            else if (RELU && bf2f(yv[i]) <= 0.f) gf = 0.f;
             if (RES) odr[i] = f2bf(gf);
             odx[i] = f2bf(gw[i] * gf);
         }
@@ -430,7 +457,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
                                   torch::Tensor running_var,
                                   bool training, double momentum, double eps,
                                   bool relu,
-                                  c10::optional<torch::Tensor> residual) {
+                                  c10::optional<torch::Tensor> residual,
+                                  bool want_mask) {
     CHECK_BN(x);
     TORCH_CHECK(x.dtype() == torch::kBFloat16, "bn_fwd: bf16 only");
     const long M = x.size(0);
@@ -494,24 +522,32 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
         CHECK_BN(residual.value());
         res_ptr = (const short*)residual.value().data_ptr();
     }
-    #define APPLY(RELU_, RES_) \
-        hipLaunchKernelGGL((bn_apply_kernel<RELU_, RES_>), agrid, \
+    auto mask = (want_mask && relu)
+        ? torch::empty({M, C / 8}, x.options().dtype(torch::kUInt8))
+        : torch::empty({0}, x.options().dtype(torch::kUInt8));
+    unsigned char* mask_ptr = (want_mask && relu)
+        ? mask.data_ptr<unsigned char>() : nullptr;
+    #define APPLY(RELU_, RES_, MASK_) \
+        hipLaunchKernelGGL((bn_apply_kernel<RELU_, RES_, MASK_>), agrid, \
                            dim3(256), 0, stream, (const short*)x.data_ptr(), \
                            res_ptr, scale_shift.data_ptr<float>(), \
-                           (short*)y.data_ptr(), M, C)
-    if (relu && res_ptr) APPLY(true, true);
-    else if (relu) APPLY(true, false);
-    else if (res_ptr) APPLY(false, true);
-    else APPLY(false, false);
+                           (short*)y.data_ptr(), mask_ptr, M, C)
+    if (mask_ptr && res_ptr) APPLY(true, true, true);
+    else if (mask_ptr) APPLY(true, false, true);
+    else if (relu && res_ptr) APPLY(true, true, false);
+    else if (relu) APPLY(true, false, false);
+    else if (res_ptr) APPLY(false, true, false);
+    else APPLY(false, false, false);
     #undef APPLY
-    return {y, save_mean, save_rstd};
+    return {y, save_mean, save_rstd, mask};
 }
 
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                   torch::Tensor x, torch::Tensor weight,
                                   torch::Tensor save_mean,
                                   torch::Tensor save_rstd,
-                                  bool training, bool relu, bool has_res) {
+                                  bool training, bool relu, bool has_res,
+                                  bool use_mask) {
     CHECK_BN(dy); CHECK_BN(x);
     const long M = x.size(0);
     const int C = x.size(1);
@@ -539,29 +575,24 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
         : dim3((int)std::max<long>(1, std::min<long>(cdiv(M, 256 / tpr), 1024)), 1);
 
     if (training) {
-        if (relu)
-            hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>),
-                               rgrid, dim3(256), 0, stream,
-                               (const short*)dy.data_ptr(),
-                               (const short*)y.data_ptr(),
-                               (const short*)x.data_ptr(),
-                               save_mean.data_ptr<float>(),
-                               save_rstd.data_ptr<float>(),
-                               partials.data_ptr<float>(), M, C);
-        else
-            hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>),
-                               rgrid, dim3(256), 0, stream,
-                               (const short*)dy.data_ptr(),
-                               (const short*)y.data_ptr(),
-                               (const short*)x.data_ptr(),
-                               save_mean.data_ptr<float>(),
-                               save_rstd.data_ptr<float>(),
-                               partials.data_ptr<float>(), M, C);
+        #define RED(RELU_, MASK_) \
+            hipLaunchKernelGGL((bn_bwd_reduce_kernel<RELU_, MASK_>), \
+                               rgrid, dim3(256), 0, stream, \
+                               (const short*)dy.data_ptr(), \
+                               (const short*)y.data_ptr(), \
+                               (const short*)x.data_ptr(), \
+                               save_mean.data_ptr<float>(), \
+                               save_rstd.data_ptr<float>(), \
+                               partials.data_ptr<float>(), M, C)
+        if (relu && use_mask) RED(true, true);
+        else if (relu) RED(true, false);
+        else RED(false, false);
+        #undef RED
         hipLaunchKernelGGL(bn_partial_sum_kernel, dim3(cdiv(2 * C, 256)),
                            dim3(256), 0, stream, partials.data_ptr<float>(),
                            sums.data_ptr<float>(), nb, 2 * C);
-        #define BWD_APPLY(RELU_, RES_) \
-            hipLaunchKernelGGL((bn_bwd_apply_kernel<RELU_, RES_>), \
+        #define BWD_APPLY(RELU_, RES_, MASK_) \
+            hipLaunchKernelGGL((bn_bwd_apply_kernel<RELU_, RES_, MASK_>), \
                                agrid, dim3(256), 0, stream, \
                                (const short*)dy.data_ptr(), \
                                (const short*)y.data_ptr(), \
@@ -572,37 +603,34 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                sums.data_ptr<float>(), \
                                (short*)dx.data_ptr(), \
                                has_res ? (short*)dres.data_ptr() : nullptr, M, C)
-        if (relu && has_res) BWD_APPLY(true, true);
-        else if (relu) BWD_APPLY(true, false);
-        else if (has_res) BWD_APPLY(false, true);
-        else BWD_APPLY(false, false);
+        if (relu && use_mask && has_res) BWD_APPLY(true, true, true);
+        else if (relu && use_mask) BWD_APPLY(true, false, true);
+        else if (relu && has_res) BWD_APPLY(true, true, false);
+        else if (relu) BWD_APPLY(true, false, false);
+        else if (has_res) BWD_APPLY(false, true, false);
+        else BWD_APPLY(false, false, false);
         #undef BWD_APPLY
     } else {
         // eval: dx = gamma*rstd*dy_eff; grads for weight/bias still need the
         // reduce (xhat uses running stats)
-        if (relu)
-            hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>),
-                               rgrid, dim3(256), 0, stream,
-                               (const short*)dy.data_ptr(),
-                               (const short*)y.data_ptr(),
-                               (const short*)x.data_ptr(),
-                               save_mean.data_ptr<float>(),
-                               save_rstd.data_ptr<float>(),
-                               partials.data_ptr<float>(), M, C);
-        else
-            hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>),
-                               rgrid, dim3(256), 0, stream,
-                               (const short*)dy.data_ptr(),
-                               (const short*)y.data_ptr(),
-                               (const short*)x.data_ptr(),
-                               save_mean.data_ptr<float>(),
-                               save_rstd.data_ptr<float>(),
-                               partials.data_ptr<float>(), M, C);
+        #define RED(RELU_, MASK_) \
+            hipLaunchKernelGGL((bn_bwd_reduce_kernel<RELU_, MASK_>), \
+                               rgrid, dim3(256), 0, stream, \
+                               (const short*)dy.data_ptr(), \
+                               (const short*)y.data_ptr(), \
+                               (const short*)x.data_ptr(), \
+                               save_mean.data_ptr<float>(), \
+                               save_rstd.data_ptr<float>(), \
+                               partials.data_ptr<float>(), M, C)
+        if (relu && use_mask) RED(true, true);
+        else if (relu) RED(true, false);
+        else RED(false, false);
+        #undef RED
         hipLaunchKernelGGL(bn_partial_sum_kernel, dim3(cdiv(2 * C, 256)),
                            dim3(256), 0, stream, partials.data_ptr<float>(),
                            sums.data_ptr<float>(), nb, 2 * C);
-        #define EVAL_APPLY(RELU_, RES_) \
-            hipLaunchKernelGGL((bn_bwd_eval_kernel<RELU_, RES_>), \
+        #define EVAL_APPLY(RELU_, RES_, MASK_) \
+            hipLaunchKernelGGL((bn_bwd_eval_kernel<RELU_, RES_, MASK_>), \
                                agrid, dim3(256), 0, stream, \
                                (const short*)dy.data_ptr(), \
                                (const short*)y.data_ptr(), \
@@ -610,10 +638,12 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                weight.data_ptr<float>(), \
                                (short*)dx.data_ptr(), \
                                has_res ? (short*)dres.data_ptr() : nullptr, M, C)
-        if (relu && has_res) EVAL_APPLY(true, true);
-        else if (relu) EVAL_APPLY(true, false);
-        else if (has_res) EVAL_APPLY(false, true);
-        else EVAL_APPLY(false, false);
+        if (relu && use_mask && has_res) EVAL_APPLY(true, true, true);
+        else if (relu && use_mask) EVAL_APPLY(true, false, true);
+        else if (relu && has_res) EVAL_APPLY(true, true, false);
+        else if (relu) EVAL_APPLY(true, false, false);
+        else if (has_res) EVAL_APPLY(false, true, false);
+        else EVAL_APPLY(false, false, false);
         #undef EVAL_APPLY
     }
     // dgamma = sum(dy_eff * xhat), dbeta = sum(dy_eff)

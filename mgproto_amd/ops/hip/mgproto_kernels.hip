@@ -535,12 +535,14 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
                                   torch::Tensor running_var,
                                   bool training, double momentum, double eps,
                                   bool relu,
-                                  c10::optional<torch::Tensor> residual);
+                                  c10::optional<torch::Tensor> residual,
+                                  bool want_mask);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                   torch::Tensor x, torch::Tensor weight,
                                   torch::Tensor save_mean,
                                   torch::Tensor save_rstd,
-                                  bool training, bool relu, bool has_res);
+                                  bool training, bool relu, bool has_res,
+                                  bool use_mask);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gmm_fwd", &gmm_fwd, "fused GMM score GEMM forward (MFMA f32)");

@@ -177,3 +177,40 @@ def test_gemm_conv1x1_parity_optin():
         assert torch.isfinite(conv.weight.grad.float()).all()
     finally:
         os.environ.pop('MGPROTO_GEMM_CONV1X1')
+
+
+@pytest.mark.skipif(os.environ.get('MGPROTO_BN_MASK') != '1',
+                    reason='experimental relu-mask BN path (round 2): '
+                           'set MGPROTO_BN_MASK=1 to validate')
+@pytest.mark.parametrize('res', [False, True])
+def test_bn_mask_backward_matches_y_path(res):
+    """EXPERIMENTAL: mask-based backward must be bitwise-identical to the
+    y-read backward (the mask is derived from the same rounded y)."""
+    from mgproto_amd.models.fused_bn import bn_act
+    torch.manual_seed(3)
+    C = 64
+    bn1 = nn.BatchNorm2d(C).cuda().train()
+    bn2 = nn.BatchNorm2d(C).cuda().train()
+    bn2.load_state_dict(bn1.state_dict())
+    x, r = _pair(C=C, seed=4, res=res)
+    g = None
+    grads = []
+    for bn, mask in ((bn1, '0'), (bn2, '1')):
+        os.environ['MGPROTO_BN_MASK'] = mask
+        try:
+            x1 = x.clone().requires_grad_(True)
+            r1 = r.clone().requires_grad_(True) if res else None
+            y = bn_act(x1, bn, relu=True, residual=r1)
+            if g is None:
+                g = torch.randn_like(y)
+            y.backward(g)
+            grads.append((x1.grad.clone(),
+                          r1.grad.clone() if res else None,
+                          bn.weight.grad.clone(), bn.bias.grad.clone()))
+        finally:
+            os.environ['MGPROTO_BN_MASK'] = '1'
+    assert torch.equal(grads[0][0], grads[1][0])
+    assert torch.equal(grads[0][2], grads[1][2])
+    assert torch.equal(grads[0][3], grads[1][3])
+    if res:
+        assert torch.equal(grads[0][1], grads[1][1])
