@@ -8,6 +8,7 @@ the parameters the reference passes.
 """
 
 import math
+import os
 import random
 from typing import Sequence, Tuple
 
@@ -18,6 +19,22 @@ try:
     from PIL import Image, ImageEnhance
 except ImportError:  # pragma: no cover
     Image = None
+
+
+_FASTAUG = [None, False]   # [module, checked]
+
+
+def _fastaug():
+    """Native warp+jitter+normalize core (mgproto_amd/ops/cpu) or None."""
+    if not _FASTAUG[1]:
+        _FASTAUG[1] = True
+        if os.environ.get('MGPROTO_NO_FASTAUG') != '1':
+            try:
+                from ..ops.hip_loader import load_cpu
+                _FASTAUG[0] = load_cpu()
+            except Exception:  # noqa: BLE001
+                _FASTAUG[0] = None
+    return _FASTAUG[0]
 
 
 class Compose:
@@ -341,10 +358,25 @@ class FusedTrainTransform:
 
         # one resample at the output size
         c = (M / M[2, 2]).reshape(9)[:8]
-        out = img.transform((S, S), Image.PERSPECTIVE, tuple(c), Image.BILINEAR)
+        params = self.jitter.sample() if self.jitter is not None \
+            else (1.0, 1.0, 1.0, 0.0, 0)
 
+        native = _fastaug()
+        if native is not None:
+            arr = torch.from_numpy(np.asarray(img.convert('RGB'),
+                                              dtype=np.uint8).copy())
+            nm = (self.normalize.mean.flatten() if self.normalize is not None
+                  else torch.zeros(3))
+            ns = (self.normalize.std.flatten() if self.normalize is not None
+                  else torch.ones(3))
+            return native.warp_jitter_normalize(
+                arr, torch.from_numpy(np.asarray(c, dtype=np.float64)), S,
+                float(params[0]), float(params[1]), float(params[2]),
+                float(params[3]), int(params[4]), nm, ns)
+
+        out = img.transform((S, S), Image.PERSPECTIVE, tuple(c), Image.BILINEAR)
         if self.jitter is not None:
-            out = self.jitter(out)
+            out = self.jitter.apply(out, params)
         t = ToTensor()(out)
         if self.normalize is not None:
             t = self.normalize(t)
@@ -364,39 +396,40 @@ class FastColorJitter:
         self.saturation = cj.saturation
         self.hue = cj.hue
 
-    def __call__(self, img):
+    def sample(self):
+        """(brightness, contrast, saturation, hue_shift, order_code 0..5)."""
+        bf = random.uniform(*self.brightness) if self.brightness else 1.0
+        cf = random.uniform(*self.contrast) if self.contrast else 1.0
+        sf = random.uniform(*self.saturation) if self.saturation else 1.0
+        hs = random.uniform(*self.hue) if self.hue else 0.0
+        return bf, cf, sf, hs, random.randrange(6)
+
+    _ORDERS = [(0, 1, 2), (0, 2, 1), (1, 0, 2), (1, 2, 0), (2, 0, 1),
+               (2, 1, 0)]
+
+    def apply(self, img, params):
+        """Apply sampled params (b/c/s in the drawn order, hue last — the
+        same convention as the native fastaug core)."""
+        bf, cf, sf, hs, order = params
         arr = np.asarray(img, dtype=np.float32)
-        ops = []
-        if self.brightness:
-            f = random.uniform(*self.brightness)
-            ops.append(lambda a: a * f)
-        if self.contrast:
-            f = random.uniform(*self.contrast)
-
-            def _ct(a, f=f):
-                m = (a @ np.array([0.299, 0.587, 0.114],
-                                  dtype=np.float32)).mean()
-                return m + f * (a - m)
-            ops.append(_ct)
-        if self.saturation:
-            f = random.uniform(*self.saturation)
-
-            def _sat(a, f=f):
-                g = a @ np.array([0.299, 0.587, 0.114], dtype=np.float32)
-                return g[..., None] + f * (a - g[..., None])
-            ops.append(_sat)
-        if self.hue:
-            shift = random.uniform(*self.hue)
-
-            def _hue(a, shift=shift):
-                im = Image.fromarray(np.clip(a, 0, 255).astype(np.uint8))
-                hsv = np.asarray(im.convert('HSV'), dtype=np.uint8).copy()
-                hsv[:, :, 0] = (hsv[:, :, 0].astype(np.int16)
-                                + int(shift * 255)) % 256
-                return np.asarray(Image.fromarray(hsv, 'HSV').convert('RGB'),
-                                  dtype=np.float32)
-            ops.append(_hue)
-        random.shuffle(ops)
-        for op in ops:
-            arr = op(arr)
+        lw = np.array([0.299, 0.587, 0.114], dtype=np.float32)
+        for op in self._ORDERS[order]:
+            if op == 0 and bf != 1.0:
+                arr = arr * bf
+            elif op == 1 and cf != 1.0:
+                m = (arr @ lw).mean()
+                arr = m + cf * (arr - m)
+            elif op == 2 and sf != 1.0:
+                g = arr @ lw
+                arr = g[..., None] + sf * (arr - g[..., None])
+        if hs != 0.0:
+            im = Image.fromarray(np.clip(arr, 0, 255).astype(np.uint8))
+            hsv = np.asarray(im.convert('HSV'), dtype=np.uint8).copy()
+            hsv[:, :, 0] = (hsv[:, :, 0].astype(np.int16)
+                            + int(hs * 255)) % 256
+            arr = np.asarray(Image.fromarray(hsv, 'HSV').convert('RGB'),
+                             dtype=np.float32)
         return Image.fromarray(np.clip(arr, 0, 255).astype(np.uint8))
+
+    def __call__(self, img):
+        return self.apply(img, self.sample())

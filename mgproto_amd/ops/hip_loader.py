@@ -10,8 +10,10 @@ import os
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
 SO_PATH = os.path.join(_HERE, '_mgproto_hip.so')
+CPU_SO_PATH = os.path.join(_HERE, '_mgproto_cpu.so')
 
 _mod = None
+_cpu_mod = None
 
 
 def load():
@@ -26,3 +28,18 @@ def load():
     spec.loader.exec_module(mod)
     _mod = mod
     return _mod
+
+
+def load_cpu():
+    """Native CPU augmentation core (optional; python fallback exists)."""
+    global _cpu_mod
+    if _cpu_mod is not None:
+        return _cpu_mod
+    if not os.path.isfile(CPU_SO_PATH):
+        raise FileNotFoundError(f'{CPU_SO_PATH} not built')
+    import torch  # noqa: F401
+    spec = importlib.util.spec_from_file_location('_mgproto_cpu', CPU_SO_PATH)
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    _cpu_mod = mod
+    return _cpu_mod

@@ -162,3 +162,49 @@ def test_fast_color_jitter_matches_pil_enhance():
     got = np.asarray(fj(img), dtype=np.float32)
     want = np.asarray(ImageEnhance.Color(img).enhance(0.5), dtype=np.float32)
     assert np.abs(got - want).mean() < 2.0
+
+
+def test_native_fastaug_matches_python_path():
+    """The C++ warp+jitter+normalize core vs the PIL/numpy path with the
+    SAME sampled parameters (the native path keeps float precision where
+    the python path quantizes to uint8 between stages — small tolerance)."""
+    import random
+    from mgproto_amd.data.preprocess import mean, std
+    if T._fastaug() is None:
+        pytest.skip('native fastaug not built')
+    img = _img(120, 100, seed=11)
+    tf = T.FusedTrainTransform(64, normalize=T.Normalize(mean, std))
+    for seed in (0, 1, 2):
+        random.seed(seed)
+        a = tf(img)
+        os.environ['MGPROTO_NO_FASTAUG'] = '1'
+        T._FASTAUG[:] = [None, False]
+        try:
+            random.seed(seed)
+            b = tf(img)
+        finally:
+            os.environ.pop('MGPROTO_NO_FASTAUG')
+            T._FASTAUG[:] = [None, False]
+        d = (a - b).abs()
+        assert float(d.mean()) < 0.08, float(d.mean())
+
+
+def test_native_fastaug_warp_only_tight():
+    """No jitter: the two paths differ only by the python path's uint8
+    round-trip — must agree to ~1 gray level."""
+    import random
+    if T._fastaug() is None:
+        pytest.skip('native fastaug not built')
+    img = _img(100, 100, seed=13)
+    tf = T.FusedTrainTransform(64, jitter=None)
+    random.seed(5)
+    a = tf(img)
+    os.environ['MGPROTO_NO_FASTAUG'] = '1'
+    T._FASTAUG[:] = [None, False]
+    try:
+        random.seed(5)
+        b = tf(img)
+    finally:
+        os.environ.pop('MGPROTO_NO_FASTAUG')
+        T._FASTAUG[:] = [None, False]
+    assert float((a - b).abs().mean()) < 0.01
