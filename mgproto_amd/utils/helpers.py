@@ -28,6 +28,12 @@ def make_one_hot(target, target_one_hot):
     target_one_hot.scatter_(dim=1, index=target, value=1.)
 
 
+def print_and_write(s, file):
+    """Reference helpers.py:33-35."""
+    print(s)
+    file.write(s + '\n')
+
+
 def makedir(path):
     if not os.path.exists(path):
         os.makedirs(path, exist_ok=True)

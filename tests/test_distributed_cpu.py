@@ -264,3 +264,7 @@ def impl_metric_allreduce(rank, world):
 
 def test_distributed_push_three_ranks():
     _run_workers(impl_distributed_push_matches_single, world=3)
+
+
+def test_grad_reducer_four_ranks():
+    _run_workers(impl_grad_reducer_matches_mean, world=4)
