@@ -36,12 +36,18 @@ def make_dataset(directory: str, class_to_idx: dict) -> List[Tuple[str, int]]:
 
 
 class ImageFolder(Dataset):
-    def __init__(self, root: str, transform: Optional[Callable] = None):
+    def __init__(self, root: str, transform: Optional[Callable] = None,
+                 decode_size: Optional[int] = None):
         self.root = root
         self.classes, self.class_to_idx = find_classes(root)
         self.samples = make_dataset(root, self.class_to_idx)
         self.imgs = self.samples  # torchvision-compat alias
         self.transform = transform
+        # When set, JPEGs are decoded by libjpeg at the smallest 1/1..1/8
+        # scale that keeps both dimensions >= decode_size (PIL draft()) —
+        # skipping most of the IDCT work for large sources. Opt-in: it
+        # slightly reduces the resolution augmentations sample from.
+        self.decode_size = decode_size
 
     def __len__(self):
         return len(self.samples)
@@ -49,7 +55,10 @@ class ImageFolder(Dataset):
     def loader(self, path):
         from PIL import Image
         with open(path, 'rb') as f:
-            return Image.open(f).convert('RGB')
+            img = Image.open(f)
+            if self.decode_size and img.format == 'JPEG':
+                img.draft('RGB', (self.decode_size, self.decode_size))
+            return img.convert('RGB')
 
     def __getitem__(self, index):
         path, target = self.samples[index]

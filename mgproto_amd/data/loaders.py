@@ -28,8 +28,12 @@ def _shard(ds, world, rank):
 
 
 def build_image_loaders(cfg, world: int = 1, rank: int = 0,
-                        fast_augment: bool = True) -> Tuple:
+                        fast_augment: bool = True,
+                        scaled_decode: bool = None) -> Tuple:
     img_size = cfg.img_size
+    if scaled_decode is None:
+        import os
+        scaled_decode = os.environ.get('MGPROTO_SCALED_DECODE', '0') == '1'
     normalize = T.Normalize(mean=mean, std=std)
 
     if fast_augment:
@@ -64,7 +68,10 @@ def build_image_loaders(cfg, world: int = 1, rank: int = 0,
         normalize,
     ])
 
-    train_ds = ImageFolder(cfg.train_dir, train_tf)
+    # Train only: 2x oversampling headroom is kept (min crop scale 0.6 of a
+    # 2*S decode still exceeds S); push/test keep full-fidelity decodes.
+    train_ds = ImageFolder(cfg.train_dir, train_tf,
+                           decode_size=2 * img_size if scaled_decode else None)
     push_ds = ImageFolder(cfg.train_push_dir, push_tf)
     test_ds = ImageFolder(cfg.test_dir, test_tf)
 

@@ -208,3 +208,32 @@ def test_native_fastaug_warp_only_tight():
         os.environ.pop('MGPROTO_NO_FASTAUG')
         T._FASTAUG[:] = [None, False]
     assert float((a - b).abs().mean()) < 0.01
+
+
+def test_scaled_jpeg_decode(tmp_path):
+    """decode_size triggers libjpeg 1/n decode for large JPEGs, and the
+    fused transform still yields the right output shape."""
+    from PIL import Image as PILImage
+    d = tmp_path / 'c0'
+    d.mkdir()
+    big = PILImage.fromarray(
+        np.random.default_rng(0).integers(0, 255, (900, 1200, 3),
+                                          dtype=np.uint8))
+    big.save(str(d / 'a.jpg'), quality=90)
+
+    tf = T.FusedTrainTransform(224, normalize=T.Normalize([0.5] * 3,
+                                                          [0.5] * 3))
+    ds = ImageFolder(str(tmp_path), tf, decode_size=448)
+    path, _ = ds.samples[0]
+    img = ds.loader(path)
+    # draft picks the smallest 1/n scale with both dims >= 448: 1/2 here
+    assert min(img.size) >= 448 and max(img.size) < 1200
+    out, label, idx = ds[0]
+    assert out.shape == (3, 224, 224) and label == 0 and idx == 0
+
+    # PNGs (no draft support) decode at full size
+    png = tmp_path / 'c0' / 'b.png'
+    big.save(str(png))
+    ds2 = ImageFolder(str(tmp_path), None, decode_size=448)
+    p2 = [p for p, _ in ds2.samples if p.endswith('.png')][0]
+    assert ds2.loader(p2).size == (1200, 900)
