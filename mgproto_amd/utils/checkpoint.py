@@ -23,6 +23,9 @@ def save_model_w_condition(model, model_dir, model_name, accu, target_accu,
 def save_train_state(path: str, model, optimizers: Dict[str, object],
                      schedulers: Dict[str, object], epoch: int,
                      extra: Optional[dict] = None):
+    import random
+
+    import numpy as np
     state = {
         'model': model.state_dict(),
         'optimizers': {k: v.state_dict() for k, v in optimizers.items() if v is not None},
@@ -31,6 +34,8 @@ def save_train_state(path: str, model, optimizers: Dict[str, object],
         'torch_rng': torch.get_rng_state(),
         'cuda_rng': (torch.cuda.get_rng_state_all()
                      if torch.cuda.is_available() else None),
+        'py_rng': random.getstate(),
+        'np_rng': np.random.get_state(),
         'extra': extra or {},
     }
     tmp = path + '.tmp'
@@ -51,4 +56,15 @@ def load_train_state(path: str, model, optimizers: Dict[str, object] = None,
             v.load_state_dict(state['schedulers'][k])
     if state.get('torch_rng') is not None:
         torch.set_rng_state(state['torch_rng'].cpu().to(torch.uint8))
+    cuda_rng = state.get('cuda_rng')
+    if (cuda_rng is not None and torch.cuda.is_available()
+            and len(cuda_rng) == torch.cuda.device_count()):
+        torch.cuda.set_rng_state_all([t.cpu().to(torch.uint8)
+                                      for t in cuda_rng])
+    if state.get('py_rng') is not None:
+        import random
+        random.setstate(state['py_rng'])
+    if state.get('np_rng') is not None:
+        import numpy as np
+        np.random.set_state(state['np_rng'])
     return state

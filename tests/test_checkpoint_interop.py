@@ -89,3 +89,24 @@ def test_nonneg_linear_prune_arg_and_score():
                    m.prototype_covs[0].unsqueeze(0),
                    torch.full((1, 4, 1), 0.25), as_average=False)
     assert per.shape == (10,)
+
+
+def test_rng_resume(tmp_path):
+    """python/numpy/torch RNG streams continue identically after resume."""
+    import random
+    import numpy as np
+    import torch
+    from mgproto_amd.utils.checkpoint import (save_train_state,
+                                              load_train_state)
+
+    m = torch.nn.Linear(2, 2)
+    random.seed(7); np.random.seed(7); torch.manual_seed(7)
+    random.random(); np.random.random(); torch.rand(1)
+    save_train_state(str(tmp_path / 'st.pt'), m, {}, {}, epoch=3)
+    expect = (random.random(), float(np.random.random()),
+              float(torch.rand(1)))
+
+    random.seed(99); np.random.seed(99); torch.manual_seed(99)
+    st = load_train_state(str(tmp_path / 'st.pt'), m)
+    got = (random.random(), float(np.random.random()), float(torch.rand(1)))
+    assert got == expect and st['epoch'] == 3
