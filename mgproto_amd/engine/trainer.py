@@ -69,7 +69,7 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
               use_mine=False, update_GMM=False, class_specific=True,
               coefs=None, log=print, device=None, amp_dtype='bf16',
               em_runner: Optional[EMRunner] = None, metrics=None,
-              comm=None, print_every=20):
+              comm=None, print_every=20, reducer=None):
     device = device or next(_unwrap(model).parameters()).device
     m = _unwrap(model)
     start = time.time()
@@ -125,9 +125,13 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
         loss = (coefs['crs_ent'] * cross_entropy + coefs['mine'] * mine_loss
                 + coefs['aux'] * aux_loss)
         with timer.phase('backward'):
+            if reducer is not None:
+                reducer.prepare()   # arm per-step bucket state (C1)
             optimizer.zero_grad(set_to_none=True)
             loss.backward()
         with timer.phase('optimizer'):
+            if reducer is not None:
+                reducer.finalize()  # drain async all-reduces before step
             optimizer.step()
 
         # EM update (reference train_and_test.py:61-63; update_interval=1)
@@ -162,6 +166,11 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
 
     if em_runner is not None:
         em_runner.sync()
+
+    if comm is not None and comm.is_distributed:
+        # BN running stats updated from rank-local batches: re-sync so
+        # eval/push see one set of stats and ranks stay bit-identical
+        comm.broadcast_buffers(m)
 
     stats = torch.stack([n_correct, n_examples, total_ce, total_mine, total_aux])
     if comm is not None:

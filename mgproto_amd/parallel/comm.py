@@ -116,3 +116,17 @@ class Comm:
                     + list(module.named_buffers()):
                 if torch.is_tensor(p) and p.numel() > 0:
                     self.broadcast(p.data if hasattr(p, 'data') else p, src=src)
+
+    def broadcast_buffers(self, module: torch.nn.Module, src: int = 0):
+        """Re-sync buffers only (BatchNorm running stats drift with each
+        rank's local batches; gradients use batch stats so training math is
+        unaffected, but eval/push read the running stats — rank 0's win,
+        DDP's broadcast_buffers convention). Called at epoch end. Buffers
+        that are already rank-identical (memory bank) ride along --
+        idempotent, and one extra epoch-rate collective is noise."""
+        if not self.is_distributed:
+            return
+        with torch.no_grad():
+            for _, b in module.named_buffers():
+                if torch.is_tensor(b) and b.numel() > 0 and b.is_floating_point():
+                    self.broadcast(b, src=src)

@@ -268,3 +268,82 @@ def test_distributed_push_three_ranks():
 
 def test_grad_reducer_four_ranks():
     _run_workers(impl_grad_reducer_matches_mean, world=4)
+
+
+def test_engine_epoch_ranks_stay_identical():
+    """Full engine epoch (tnt.train with reducer + DistributedEnqueue + EM)
+    on 2 ranks with DIFFERENT data shards: all model/aux parameters and the
+    memory bank must end bit-identical across ranks. Regression test for
+    the reducer not being armed per step inside the engine loop."""
+    res = _run_workers(impl_engine_epoch_ranks_stay_identical, world=2)
+    for k in res[0]['digests']:
+        assert res[0]['digests'][k] == res[1]['digests'][k], \
+            f'rank drift in {k}'
+    # different shards must actually move the weights (reduction is not a
+    # no-op freeze)
+    assert res[0]['moved'] > 0
+
+
+def impl_engine_epoch_ranks_stay_identical(rank, world):
+    from torch.utils.data import DataLoader
+
+    from mgproto_amd.model import construct_MGProto
+    from mgproto_amd.data import SyntheticImages
+    from mgproto_amd.engine import joint
+    from mgproto_amd.engine import train as tnt_train
+    from mgproto_amd.losses import build_aux_loss
+    from mgproto_amd.parallel import (Comm, BucketedGradReducer,
+                                      make_dp_correct)
+
+    comm = Comm(backend='gloo')
+    C, K, d = 6, 2, 16
+    torch.manual_seed(100 + rank)   # different init per rank on purpose:
+    model = construct_MGProto('resnet34', pretrained=False, img_size=64,
+                              prototype_shape=(C * K, d, 1, 1),
+                              num_classes=C, add_on_layers_type='regular',
+                              sz_embedding=8, mem_capacity=8, mine_K=2)
+    aux = build_aux_loss('Proxy_Anchor', nb_classes=C, sz_embed=8)
+    # ... make_dp_correct broadcasts rank 0's state
+    model = make_dp_correct(model, comm, train_batch_size=4)
+    comm.broadcast_module(aux)
+    reducer = BucketedGradReducer([model, aux], comm, bucket_mb=1)
+    opt = torch.optim.Adam([
+        {'params': model.parameters(), 'lr': 1e-3},
+        {'params': aux.parameters(), 'lr': 1e-3},
+    ])
+    joint(model)
+    before = model.add_on_layers[0].weight.detach().clone()
+
+    def _collate(batch):
+        return (torch.stack([b[0] for b in batch]),
+                torch.tensor([b[1] for b in batch]),
+                torch.tensor([b[2] for b in batch]))
+
+    ds = SyntheticImages(n=16, num_classes=C, img_size=64, seed=500 + rank)
+    loader = DataLoader(ds, batch_size=4, collate_fn=_collate)
+    # prefill the bank so the EM path runs this epoch (rank-invariant, as
+    # DistributedEnqueue guarantees for the in-training enqueue path)
+    g = torch.Generator().manual_seed(4242)
+    for c in range(C):
+        model.queue.push(torch.nn.functional.normalize(
+            torch.randn(8, d, generator=g), dim=1),
+            torch.full((8,), c, dtype=torch.long))
+        model.memory_updated_cls[c] = True
+
+    tnt_train(model, loader, opt, aux_criterion=aux, use_mine=True,
+              update_GMM=True, coefs={'crs_ent': 1, 'mine': 0.2, 'aux': 0.5},
+              log=lambda *a: None, amp_dtype='off', print_every=0,
+              comm=comm, reducer=reducer)
+
+    import hashlib
+
+    def digest(t):
+        return hashlib.sha256(
+            t.detach().cpu().contiguous().float().numpy().tobytes()).hexdigest()
+
+    digests = {f'model/{k}': digest(v) for k, v in model.state_dict().items()}
+    digests.update({f'aux/{k}': digest(v)
+                    for k, v in aux.state_dict().items()})
+    moved = float((model.add_on_layers[0].weight.detach()
+                   - before).abs().sum())
+    return {'digests': digests, 'moved': moved}

@@ -233,7 +233,7 @@ def main():
             _, train_results = tnt.train(
                 ppnet, train_loader, warm_optimizer, aux_criterion=aux_criterion,
                 use_mine=use_mining, update_GMM=update_GMM, coefs=cfg.coefs,
-                log=log, em_runner=em_runner, **kw)
+                log=log, em_runner=em_runner, reducer=reducer, **kw)
         else:
             tnt.joint(ppnet, log=log)
             if epoch in decay_epochs:
@@ -241,7 +241,7 @@ def main():
             _, train_results = tnt.train(
                 ppnet, train_loader, joint_optimizer, aux_criterion=aux_criterion,
                 use_mine=use_mining, update_GMM=update_GMM, coefs=cfg.coefs,
-                log=log, em_runner=em_runner, **kw)
+                log=log, em_runner=em_runner, reducer=reducer, **kw)
 
         if args.ood_eval:
             accu, _ = tnt.test(ppnet, (test_loader, ood1_loader, ood2_loader),
