@@ -30,19 +30,32 @@ def build_argparser():
     p.add_argument('--resume', type=str, required=True)
     p.add_argument('--img_size', type=int, default=224)
     p.add_argument('--half_size', type=int, default=36)
+    p.add_argument('--mem_sz', type=int, default=None,
+                   help='memory-bank capacity per class; default: inferred '
+                        'from the checkpoint')
     return p
 
 
 def load_model(args, device):
     P, d, ks = args.prototype_shape
+    sd = torch.load(args.resume, map_location='cpu', weights_only=False)
+    if 'model' in sd:
+        sd = sd['model']
+    # shape-bearing hyperparams not on the reference CLI: infer from the
+    # checkpoint so any training config loads
+    kw = {}
+    mem_sz = args.mem_sz
+    if mem_sz is None and 'queue.cls0' in sd:
+        mem_sz = sd['queue.cls0'].shape[0]
+    if mem_sz is not None:
+        kw['mem_capacity'] = mem_sz
+    if 'embedding.weight' in sd:
+        kw['sz_embedding'] = sd['embedding.weight'].shape[0]
     model = construct_MGProto(args.base_architecture, pretrained=False,
                               img_size=args.img_size,
                               prototype_shape=(P, d, ks, ks),
                               num_classes=args.nb_classes,
-                              add_on_layers_type=args.addon)
-    sd = torch.load(args.resume, map_location='cpu', weights_only=False)
-    if 'model' in sd:
-        sd = sd['model']
+                              add_on_layers_type=args.addon, **kw)
     model.load_state_dict(sd, strict=False)
     return model.to(device).eval()
 

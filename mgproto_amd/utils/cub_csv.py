@@ -60,12 +60,24 @@ def _sweep(net, loader, device):
     return torch.cat(pooled_all), torch.cat(hw_all), wshape
 
 
+def _dataset_imgs(ds):
+    """(path, target) list for ImageFolder-style datasets (.imgs, as the
+    reference's project loaders provide) or Cub2011Eval (metadata frame)."""
+    if hasattr(ds, 'imgs'):
+        return ds.imgs
+    if hasattr(ds, 'data') and hasattr(ds, 'root'):
+        return [(os.path.join(ds.root, 'images', r.filepath), r.target - 1)
+                for _, r in ds.data.iterrows()]
+    raise TypeError(f'unsupported dataset type {type(ds).__name__}: '
+                    'need .imgs or CUB metadata')
+
+
 @torch.no_grad()
 def get_topk_cub(net, projectloader, k, epoch, device, log_dir, img_size=224):
     """Write the top-k patch-coordinate CSV per prototype
     (reference cub_csv.py:267-349)."""
     m = net.module if hasattr(net, 'module') else net
-    imgs = projectloader.dataset.imgs
+    imgs = _dataset_imgs(projectloader.dataset)
     weights = m.last_layer.weight
     pooled, hw, wshape = _sweep(net, projectloader, device)
     patchsize, skip = get_patch_size(img_size, wshape)
@@ -99,7 +111,7 @@ def get_proto_patches_cub(net, projectloader, epoch, device, log_dir,
     """Write all patches whose similarity exceeds threshold
     (reference cub_csv.py:226-265)."""
     m = net.module if hasattr(net, 'module') else net
-    imgs = projectloader.dataset.imgs
+    imgs = _dataset_imgs(projectloader.dataset)
     weights = m.last_layer.weight
     pooled, hw, wshape = _sweep(net, projectloader, device)
     patchsize, skip = get_patch_size(img_size, wshape)
