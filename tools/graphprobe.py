@@ -37,6 +37,10 @@ def try_capture(name, fn, warm=3):
 
 
 def main():
+    if not torch.cuda.is_available():
+        print('graphprobe: GPU-only diagnostic (hipGraph capture probe); '
+              'no HIP device found.')
+        return
     from mgproto_amd.utils.helpers import setup_miopen_db
     setup_miopen_db()
     torch.backends.cudnn.benchmark = True
