@@ -17,6 +17,11 @@ MI355X re-design:
   results are identical for any GPU count (SURVEY.md hard part #4);
   re-forwards are sharded round-robin and combined with one all-reduce.
 * artifact rendering uses PIL/torch (cv2/matplotlib are not available).
+
+``prototype_self_act_filename_prefix`` / ``proto_bound_boxes_filename_prefix``
+are accepted for signature parity but have no effect — same as the
+reference, where the corresponding arrays are allocated (push.py:63-67)
+but never written or saved (SURVEY.md §2.1 dead spots).
 """
 
 import os
