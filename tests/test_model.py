@@ -240,3 +240,67 @@ def test_regular_upsample_grid():
         bf, dist = m.push_forward(torch.randn(1, 3, 64, 64))
     # 64/16 = 4 latent, x2 upsample -> 8x8 grid
     assert dist.shape[-2:] == (8, 8)
+
+
+def test_update_gmm_matches_per_class_loop():
+    """Batched masked update_GMM == an independent per-class EM loop using
+    the (separately oracle-tested) e/m-step ops + manual Adam + pi
+    momentum. Covers the masking/momentum/state bookkeeping composition:
+    only dirty+full classes move, and exactly as the naive loop says."""
+    import math
+
+    from mgproto_amd.ops import reference as R
+
+    torch.manual_seed(3)
+    C, K, d, cap = 4, 3, 8, 6
+    model = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                              prototype_shape=(C * K, d, 1, 1),
+                              num_classes=C, add_on_layers_type='regular',
+                              sz_embedding=8, mem_capacity=cap, mine_K=2)
+    # classes 0 and 2: full + dirty; class 1: full but clean; class 3:
+    # dirty but not full -> only 0 and 2 may move
+    for c in (0, 1, 2):
+        model.queue.push(F.normalize(torch.randn(cap, d), dim=1),
+                         torch.full((cap,), c, dtype=torch.long))
+    model.queue.push(F.normalize(torch.randn(2, d), dim=1),
+                     torch.full((2,), 3, dtype=torch.long))
+    model.memory_updated_cls[[0, 2, 3]] = True
+
+    w0 = model.last_layer.weight.data.clone()
+    means0 = model.prototype_means.data.clone()
+    covs = model.prototype_covs.data.clone()
+    bank = model.queue.mem.clone()
+
+    model.update_GMM()
+
+    diag = torch.arange(C)
+    pi_after = model.last_layer.weight.data.view(C, C, K)[diag, diag]
+    b1, b2 = model.adam_betas
+
+    for c in range(C):
+        if c not in (0, 2):     # clean or not-full classes: untouched
+            assert torch.equal(model.prototype_means.data[c], means0[c])
+            assert torch.equal(pi_after[c], w0.view(C, C, K)[c, c])
+            continue
+        x = bank[c:c + 1]
+        mu = means0[c:c + 1].clone()
+        pi_old = w0.view(C, C, K)[c, c].view(1, K).clone()
+        m = torch.zeros_like(mu)
+        v = torch.zeros_like(mu)
+        for step in range(1, model.num_em_loop + 1):
+            wlp, log_resp = R.em_e_step(x, mu, covs[c:c + 1], pi_old)
+            grad, pi_unnorm = R.em_m_step_grads(
+                x, log_resp, wlp, mu, covs[c:c + 1],
+                alpha=model.alpha, lamda=model.lamda)
+            m = b1 * m + (1 - b1) * grad
+            v = b2 * v + (1 - b2) * grad * grad
+            mu = mu - model.prototype_lr * (m / (1 - b1 ** step)) \
+                / ((v / (1 - b2 ** step)).sqrt() + model.adam_eps)
+            pi = pi_unnorm / cap
+            pi_old = model.tau * pi_old + (1 - model.tau) * pi
+        assert torch.allclose(model.prototype_means.data[c], mu[0],
+                              atol=1e-5), c
+        assert torch.allclose(pi_after[c], pi_old[0], atol=1e-6), c
+
+    # dirty flags consumed
+    assert not model.memory_updated_cls.any()
