@@ -128,3 +128,38 @@ def test_em_grads_match_autograd_fuzz(G, K, seed):
         (wll + div).backward()
         assert torch.allclose(grad[gi], mu.grad, atol=1e-4, rtol=1e-3), \
             (grad[gi] - mu.grad).abs().max()
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(min_value=16, max_value=96),
+       st.integers(min_value=16, max_value=96),
+       st.integers(min_value=0, max_value=10_000))
+def test_fastaug_native_vs_python_fuzz(H, W, seed):
+    """Native C++ warp core vs the PIL fallback across random image sizes
+    and random geometric draws (jitter off -> only the uint8 round-trip
+    separates them)."""
+    import os
+    import random
+
+    import numpy as np
+    from PIL import Image
+
+    from mgproto_amd.data import transforms as T
+
+    if T._fastaug() is None:
+        import pytest
+        pytest.skip('native fastaug not built')
+    rng = np.random.default_rng(seed)
+    img = Image.fromarray(rng.integers(0, 255, (H, W, 3), dtype=np.uint8))
+    tf = T.FusedTrainTransform(32, jitter=None)
+    random.seed(seed)
+    a = tf(img)
+    os.environ['MGPROTO_NO_FASTAUG'] = '1'
+    T._FASTAUG[:] = [None, False]
+    try:
+        random.seed(seed)
+        b = tf(img)
+    finally:
+        os.environ.pop('MGPROTO_NO_FASTAUG')
+        T._FASTAUG[:] = [None, False]
+    assert float((a - b).abs().mean()) < 0.02
