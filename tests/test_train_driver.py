@@ -73,3 +73,20 @@ def test_train_driver_two_ranks(tmp_path):
     assert (out_dir / 'latest.pth').is_file()
     recs = [json.loads(l) for l in open(out_dir / 'metrics.jsonl')]
     assert any('test/acc' in r for r in recs)
+
+
+def test_auto_resume_elastic_restart(tmp_path):
+    """A relaunch with the same --out continues from the newest checkpoint
+    (elastic-restart semantics); --no-auto-resume starts fresh."""
+    _run_train(tmp_path)                                   # epoch 0
+    r = _run_train(tmp_path, extra=('--epochs', '2'))      # relaunch
+    log_text = open(tmp_path / 'run' / 'train.log').read()
+    assert 'resumed from' in log_text
+    state = torch.load(tmp_path / 'run' / 'latest.pth', map_location='cpu',
+                       weights_only=False)
+    assert state['epoch'] == 1                             # continued, not redone
+
+    r = _run_train(tmp_path, extra=('--epochs', '1', '--no-auto-resume'))
+    state = torch.load(tmp_path / 'run' / 'latest.pth', map_location='cpu',
+                       weights_only=False)
+    assert state['epoch'] == 0                             # fresh run

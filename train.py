@@ -87,6 +87,11 @@ def main():
     parser.add_argument('--out', type=str, default=None)
     parser.add_argument('--ood-eval', action='store_true')
     parser.add_argument('--addon', type=str, default=None)
+    parser.add_argument('--no-auto-resume', action='store_true',
+                        help='do not resume from an existing latest.pth / '
+                             'preempt.pth in --out (elastic-restart default: '
+                             'resume, so a torchrun relaunch after preemption '
+                             'continues the run)')
     parser.add_argument('--seed', type=int, default=None,
                         help='deterministic init/data seeding (the reference '
                              'ships seeding commented out, main.py:45-49)')
@@ -187,13 +192,22 @@ def main():
     em_runner = EMRunner(ppnet, use_stream=(device.type == 'cuda'
                                             and cfg.em_stream))
     start_epoch = 0
-    if args.resume:
-        state = load_train_state(args.resume, ppnet,
+    resume_path = args.resume
+    if resume_path is None and not args.no_auto_resume:
+        # elastic restart: a relaunch after preemption/crash picks up the
+        # newest checkpoint in the run dir automatically
+        cands = [os.path.join(model_dir, f)
+                 for f in ('latest.pth', 'preempt.pth')]
+        cands = [p for p in cands if os.path.isfile(p)]
+        if cands:
+            resume_path = max(cands, key=os.path.getmtime)
+    if resume_path:
+        state = load_train_state(resume_path, ppnet,
                                  {'joint': joint_optimizer, 'warm': warm_optimizer},
                                  {'joint_lr': joint_lr_scheduler},
                                  map_location=device)
         start_epoch = state['epoch'] + 1
-        log(f'resumed from {args.resume} at epoch {start_epoch}')
+        log(f'resumed from {resume_path} at epoch {start_epoch}')
 
     # failure handling: checkpoint on SIGTERM/SIGUSR1 (preemption-safe;
     # the reference loses everything past the last conditional save)
