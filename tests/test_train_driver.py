@@ -90,3 +90,21 @@ def test_auto_resume_elastic_restart(tmp_path):
     state = torch.load(tmp_path / 'run' / 'latest.pth', map_location='cpu',
                        weights_only=False)
     assert state['epoch'] == 0                             # fresh run
+
+
+def test_train_driver_seed_determinism(tmp_path):
+    """Two identical seeded runs end with bit-identical model state."""
+    import hashlib
+
+    def digest(p):
+        sd = torch.load(p, map_location='cpu', weights_only=False)['model']
+        h = hashlib.sha256()
+        for k in sorted(sd):
+            h.update(k.encode())
+            h.update(sd[k].float().contiguous().numpy().tobytes())
+        return h.hexdigest()
+
+    _run_train(tmp_path / 'a', extra=('--seed', '7', '--no-auto-resume'))
+    _run_train(tmp_path / 'b', extra=('--seed', '7', '--no-auto-resume'))
+    assert digest(tmp_path / 'a' / 'run' / 'latest.pth') \
+        == digest(tmp_path / 'b' / 'run' / 'latest.pth')
