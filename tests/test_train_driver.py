@@ -108,3 +108,9 @@ def test_train_driver_seed_determinism(tmp_path):
     _run_train(tmp_path / 'b', extra=('--seed', '7', '--no-auto-resume'))
     assert digest(tmp_path / 'a' / 'run' / 'latest.pth') \
         == digest(tmp_path / 'b' / 'run' / 'latest.pth')
+
+
+def test_train_driver_regular_upsample(tmp_path):
+    """Driver run with the flagship add-on (convs-then-upsample, 2x grid)."""
+    _run_train(tmp_path, extra=('--addon', 'regular_upsample',))
+    assert (tmp_path / 'run' / 'latest.pth').is_file()
