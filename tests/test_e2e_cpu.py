@@ -5,6 +5,8 @@ Uses a reduced image size to keep runtime sane; the full-size path is
 exercised on GPU (test_gpu_e2e.py).
 """
 
+import os
+
 import torch
 from torch.utils.data import DataLoader
 
@@ -123,3 +125,30 @@ def test_density_auroc():
     assert _density_auroc(ident, ident) == 0.5 or \
         abs(_density_auroc(ident, ident) - 0.5) < 0.2
     assert _density_auroc(torch.zeros(0), ood) is None
+
+
+def test_push_renders_artifacts(tmp_path):
+    """Push writes the reference's three JPEG artifacts per pushed
+    prototype (original+bbox, heatmap overlay, cropped patch —
+    reference push.py:203-226)."""
+    torch.manual_seed(1)
+    C, K, d = 4, 2, 16
+    model = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                              prototype_shape=(C * K, d, 1, 1), num_classes=C,
+                              add_on_layers_type='regular', sz_embedding=8,
+                              mem_capacity=8, mine_K=2)
+    ds = SyntheticImages(n=12, num_classes=C, img_size=64, normalize=False)
+    loader = DataLoader(ds, batch_size=4, collate_fn=_collate)
+    chosen = push_prototypes(loader, model,
+                             root_dir_for_saving_prototypes=str(tmp_path),
+                             epoch_number=3, log=lambda *a: None)
+    assert chosen
+    out = tmp_path / 'epoch-3'
+    files = sorted(os.listdir(out))
+    j = chosen[0][0]
+    for suffix in ('-original.jpg', '-original_with_self_act.jpg', '.jpg'):
+        assert f'{j}prototype-img{suffix}' in files, (suffix, files[:6])
+    # artifacts decode as images
+    from PIL import Image
+    with Image.open(out / f'{j}prototype-img-original.jpg') as im:
+        assert im.size == (64, 64)
