@@ -347,3 +347,71 @@ def impl_engine_epoch_ranks_stay_identical(rank, world):
     moved = float((model.add_on_layers[0].weight.detach()
                    - before).abs().sum())
     return {'digests': digests, 'moved': moved}
+
+
+def test_ood_eval_two_ranks_matches_single():
+    """_testing_with_OoD on 2 rank-sharded loaders == single-process run
+    over the full set (quantile threshold and FPR are order-invariant)."""
+    res = _run_workers(impl_ood_eval_two_ranks, world=2)
+    assert res[0] == res[1]                      # rank-invariant outputs
+    single = impl_ood_eval_single()
+    assert abs(res[0]['acc'] - single['acc']) < 1e-6
+    assert abs(res[0]['FPR95_1'] - single['FPR95_1']) < 1e-6
+
+
+def _ood_model_and_data():
+    from torch.utils.data import DataLoader
+
+    from mgproto_amd.data import SyntheticImages
+    from mgproto_amd.model import construct_MGProto
+
+    torch.manual_seed(11)
+    C, K, d = 4, 2, 16
+    model = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                              prototype_shape=(C * K, d, 1, 1),
+                              num_classes=C, add_on_layers_type='regular',
+                              sz_embedding=8, mem_capacity=8, mine_K=2)
+    model.eval()
+
+    def collate(batch):
+        return (torch.stack([b[0] for b in batch]),
+                torch.tensor([b[1] for b in batch]),
+                torch.tensor([b[2] for b in batch]))
+
+    id_ds = SyntheticImages(n=16, num_classes=C, img_size=64, seed=5)
+    ood_ds = SyntheticImages(n=8, num_classes=C, img_size=64, seed=9)
+    return model, id_ds, ood_ds, collate, DataLoader
+
+
+def impl_ood_eval_two_ranks(rank, world):
+    from torch.utils.data import Subset
+
+    from mgproto_amd.engine import _testing_with_OoD
+    from mgproto_amd.parallel import Comm
+
+    comm = Comm(backend='gloo')
+    model, id_ds, ood_ds, collate, DataLoader = _ood_model_and_data()
+    comm.broadcast_module(model)
+
+    def shard(ds):
+        return Subset(ds, list(range(rank, len(ds), world)))
+
+    id_loader = DataLoader(shard(id_ds), batch_size=4, collate_fn=collate)
+    ood_loader = DataLoader(shard(ood_ds), batch_size=4, collate_fn=collate)
+    _, results = _testing_with_OoD(model, (id_loader, ood_loader),
+                                   log=lambda *a: None, amp_dtype='off',
+                                   comm=comm)
+    return {'acc': float(results['acc']),
+            'FPR95_1': float(results['FPR95_1'])}
+
+
+def impl_ood_eval_single():
+    from mgproto_amd.engine import _testing_with_OoD
+
+    model, id_ds, ood_ds, collate, DataLoader = _ood_model_and_data()
+    id_loader = DataLoader(id_ds, batch_size=4, collate_fn=collate)
+    ood_loader = DataLoader(ood_ds, batch_size=4, collate_fn=collate)
+    _, results = _testing_with_OoD(model, (id_loader, ood_loader),
+                                   log=lambda *a: None, amp_dtype='off')
+    return {'acc': float(results['acc']),
+            'FPR95_1': float(results['FPR95_1'])}
