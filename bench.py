@@ -229,7 +229,8 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
-        grid = 28 if args.addon == 'regular_upsample' else 14
+        # stride-16 trunk; the upsample add-on doubles the latent grid
+        grid = (args.img // 16) * (2 if args.addon == 'regular_upsample' else 1)
         print(json.dumps({
             'metric': 'train_images_per_sec',
             'value': round(ips, 2),
@@ -241,7 +242,7 @@ def main():
             'higher_is_better': True,
             'scaling': 'weak',
             'vs_baseline': None,
-            'dtype': 'bf16',
+            'dtype': 'bf16' if amp is not None else 'fp32',
             'data': 'synthetic',
             'config': {
                 'model': f'{args.arch}-mgproto',
