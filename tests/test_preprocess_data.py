@@ -57,3 +57,55 @@ def test_pets_restructure(tmp_path):
     _run('pets-restructure', '--root', str(imgs), '--labels', str(labels),
          '--out', str(out))
     assert (out / '3' / 'cat_1.jpg').is_file()
+
+
+def test_crop_cars(tmp_path):
+    scipy_io = pytest.importorskip('scipy.io')
+    root = tmp_path / 'cars'
+    (root / 'car_ims').mkdir(parents=True)
+    rng = np.random.RandomState(1)
+    Image.fromarray(rng.randint(0, 255, (60, 90, 3), dtype=np.uint8)) \
+        .save(root / 'car_ims' / '000001.jpg')
+    ann = np.zeros((1,), dtype=[('relative_im_path', 'O'),
+                                ('bbox_x1', 'O'), ('bbox_y1', 'O'),
+                                ('bbox_x2', 'O'), ('bbox_y2', 'O'),
+                                ('class', 'O'), ('test', 'O')])
+    ann[0] = ('car_ims/000001.jpg', 10, 5, 50, 45, 3, 0)
+    mat = tmp_path / 'cars_annos.mat'
+    scipy_io.savemat(str(mat), {'annotations': ann.reshape(1, -1)})
+    out = tmp_path / 'out'
+    _run('crop-cars', '--root', str(root), '--annos', str(mat),
+         '--out', str(out))
+    cropped = Image.open(out / 'train_cropped' / '003' / '000001.jpg')
+    assert cropped.size == (40, 40)
+
+
+def test_binarize_masks(tmp_path):
+    root = tmp_path / 'segs'
+    (root / '001.A').mkdir(parents=True)
+    # three gray levels: the two darkest are background
+    m = np.zeros((20, 20), dtype=np.uint8)
+    m[5:15, 5:15] = 128
+    m[8:12, 8:12] = 255
+    Image.fromarray(m).save(root / '001.A' / 'x.png')
+    out = tmp_path / 'out'
+    _run('binarize-masks', '--root', str(root), '--out', str(out))
+    fg = np.array(Image.open(out / '001.A' / 'x.png'))
+    assert set(np.unique(fg)) <= {0, 255}
+    assert fg[10, 10] == 255 and fg[0, 0] == 0 and fg[6, 6] == 0
+
+
+def test_crop_masks(tmp_path):
+    cub = tmp_path / 'cub'
+    cub.mkdir()
+    (cub / 'images.txt').write_text('1 001.A/x.jpg\n')
+    (cub / 'bounding_boxes.txt').write_text('1 4.0 2.0 10.0 8.0\n')
+    masks = tmp_path / 'masks'
+    (masks / '001.A').mkdir(parents=True)
+    Image.fromarray(np.full((30, 30), 255, dtype=np.uint8)) \
+        .save(masks / '001.A' / 'x.png')
+    out = tmp_path / 'out'
+    _run('crop-masks', '--root', str(masks), '--cub-root', str(cub),
+         '--out', str(out))
+    cropped = Image.open(out / '001.A' / 'x.png')
+    assert cropped.size == (10, 8)
