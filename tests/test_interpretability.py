@@ -184,3 +184,39 @@ def test_cub_csv_pipeline(synthetic_cub, tmp_path):
         os.path.join(str(synthetic_cub), 'images.txt'),
         epoch=0, img_size=64, wshape=4, log=lambda *a: None)
     assert 0.0 <= mean_p <= 1.0
+
+
+def test_cub_csv_threshold_variant(synthetic_cub, tmp_path):
+    """get_proto_patches_cub (threshold CSV, reference cub_csv.py:226-265)
+    writes scoreable rows for every patch above threshold."""
+    import csv as _csv
+
+    from torch.utils.data import DataLoader
+
+    from mgproto_amd.data import transforms as T
+    from mgproto_amd.model import construct_MGProto
+    from mgproto_amd.utils import cub_csv
+    from mgproto_amd.utils.datasets import Cub2011Eval
+
+    torch.manual_seed(0)
+    model = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                              prototype_shape=(8, 16, 1, 1), num_classes=2,
+                              add_on_layers_type='regular', sz_embedding=8,
+                              mem_capacity=4, mine_K=2)
+    tf = T.Compose([T.Resize((64, 64)), T.ToTensor()])
+    ds = Cub2011Eval(str(synthetic_cub), train=False, transform=tf)
+    loader = DataLoader(ds, batch_size=3)
+    csvpath = cub_csv.get_proto_patches_cub(
+        model, loader, epoch=0, device=torch.device('cpu'),
+        log_dir=str(tmp_path / 'logs'), threshold=0.0, img_size=64)
+    with open(csvpath, newline='') as f:
+        rows = list(_csv.reader(f))
+    assert rows[0][:2] == ['prototype', 'img name']
+    assert len(rows) > 1           # threshold 0: probs are positive
+    mean_p, _std, _rel = cub_csv.eval_prototypes_cub_parts_csv(
+        csvpath,
+        os.path.join(str(synthetic_cub), 'parts', 'part_locs.txt'),
+        os.path.join(str(synthetic_cub), 'parts', 'parts.txt'),
+        os.path.join(str(synthetic_cub), 'images.txt'),
+        epoch=0, img_size=64, wshape=4, log=lambda *a: None)
+    assert 0.0 <= mean_p <= 1.0
