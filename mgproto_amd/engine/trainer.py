@@ -142,7 +142,8 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
                     (em_runner.run() if em_runner is not None
                      else m.update_GMM())
 
-        if print_every and i % print_every == 0:
+        if print_every and i % print_every == 0 \
+                and (comm is None or comm.rank == 0):
             if timer.enabled:
                 phases = timer.summary()
                 print('  phase ms: ' + '  '.join(
