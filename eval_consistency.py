@@ -43,14 +43,10 @@ def load_model(args, device):
         sd = sd['model']
     # shape-bearing hyperparams not on the reference CLI: infer from the
     # checkpoint so any training config loads
-    kw = {}
-    mem_sz = args.mem_sz
-    if mem_sz is None and 'queue.cls0' in sd:
-        mem_sz = sd['queue.cls0'].shape[0]
-    if mem_sz is not None:
-        kw['mem_capacity'] = mem_sz
-    if 'embedding.weight' in sd:
-        kw['sz_embedding'] = sd['embedding.weight'].shape[0]
+    from mgproto_amd.utils.checkpoint import infer_ctor_kwargs_from_state
+    kw = infer_ctor_kwargs_from_state(sd)
+    if args.mem_sz is not None:
+        kw['mem_capacity'] = args.mem_sz
     model = construct_MGProto(args.base_architecture, pretrained=False,
                               img_size=args.img_size,
                               prototype_shape=(P, d, ks, ks),

@@ -20,6 +20,18 @@ def save_model_w_condition(model, model_dir, model_name, accu, target_accu,
                    f=os.path.join(model_dir, (model_name + '{0:.4f}.pth').format(accu)))
 
 
+def infer_ctor_kwargs_from_state(sd: dict) -> dict:
+    """Shape-bearing ``construct_MGProto`` kwargs that the reference CLI
+    does not expose, read off a checkpoint's state_dict so any training
+    config loads (memory capacity per class, aux embedding size)."""
+    kw = {}
+    if 'queue.cls0' in sd:
+        kw['mem_capacity'] = sd['queue.cls0'].shape[0]
+    if 'embedding.weight' in sd:
+        kw['sz_embedding'] = sd['embedding.weight'].shape[0]
+    return kw
+
+
 def save_train_state(path: str, model, optimizers: Dict[str, object],
                      schedulers: Dict[str, object], epoch: int,
                      extra: Optional[dict] = None):

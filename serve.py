@@ -40,14 +40,21 @@ def main():
         setup_miopen_db()
         torch.backends.cudnn.benchmark = True
 
+    kw = {}
+    sd = None
+    if args.resume:
+        from mgproto_amd.utils.checkpoint import infer_ctor_kwargs_from_state
+        sd = torch.load(args.resume, map_location=device, weights_only=False)
+        sd = sd.get('model', sd)
+        kw = infer_ctor_kwargs_from_state(sd)
     model = construct_MGProto(
         args.arch, pretrained=False, img_size=args.img,
         prototype_shape=(args.classes * args.proto_per_class,
                          args.proto_dim, 1, 1),
-        num_classes=args.classes, add_on_layers_type=args.addon).to(device)
-    if args.resume:
-        sd = torch.load(args.resume, map_location=device, weights_only=False)
-        model.load_state_dict(sd.get('model', sd), strict=False)
+        num_classes=args.classes, add_on_layers_type=args.addon,
+        **kw).to(device)
+    if sd is not None:
+        model.load_state_dict(sd, strict=False)
     if device.type == 'cuda':
         model.features = model.features.to(memory_format=torch.channels_last)
 

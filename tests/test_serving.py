@@ -97,3 +97,32 @@ def test_serving_graph_capture_gpu():
     for w, g in zip(want, got):
         assert w['pred_class'] == g['pred_class']
         assert abs(w['density_pX'] - g['density_pX']) < 1e-3 * (1 + abs(w['density_pX']))
+
+
+def test_serve_cli_loads_any_config_checkpoint(tmp_path):
+    """serve.py infers mem/embedding sizes from the checkpoint (same as the
+    eval drivers) — loading a non-default-config checkpoint must not crash.
+    Exercised via the loader path, not the HTTP server."""
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    from mgproto_amd.model import construct_MGProto
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(6, 16, 1, 1), num_classes=2,
+                          add_on_layers_type='regular', sz_embedding=8,
+                          mem_capacity=8, mine_K=2)
+    ckpt = tmp_path / 'm.pth'
+    torch.save(m.state_dict(), str(ckpt))
+    code = (
+        "import sys; sys.argv = ['serve.py', '--resume', %r, '--arch',"
+        " 'resnet18', '--classes', '2', '--proto-per-class', '3',"
+        " '--proto-dim', '16', '--img', '64', '--addon', 'regular'];"
+        "import unittest.mock as mock, uvicorn;"
+        "mock.patch.object(uvicorn, 'run').start();"
+        "import serve; serve.main(); print('LOADED-OK')" % str(ckpt))
+    r = subprocess.run([sys.executable, '-c', code], capture_output=True,
+                       text=True, timeout=300, cwd=root)
+    assert r.returncode == 0 and 'LOADED-OK' in r.stdout, \
+        r.stdout[-2000:] + r.stderr[-2000:]
