@@ -114,3 +114,10 @@ def test_train_driver_regular_upsample(tmp_path):
     """Driver run with the flagship add-on (convs-then-upsample, 2x grid)."""
     _run_train(tmp_path, extra=('--addon', 'regular_upsample',))
     assert (tmp_path / 'run' / 'latest.pth').is_file()
+
+
+def test_train_driver_ood_eval(tmp_path):
+    """--ood-eval evaluates FPR95/AUROC against the synthetic OoD sets."""
+    _run_train(tmp_path, extra=('--ood-eval',))
+    log_text = open(tmp_path / 'run' / 'train.log').read()
+    assert 'FPR95_1' in log_text and 'FPR95_2' in log_text
