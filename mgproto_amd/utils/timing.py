@@ -1,11 +1,13 @@
 """Per-phase step timing (observability; SURVEY.md §5 tracing).
 
-CUDA-event based phase timers with negligible overhead when disabled.
-Enable with ``MGPROTO_TIMING=1``; the trainer then logs per-phase means
-every print interval. For kernel-level data use rocprofv3 (profiles/).
+CUDA-event based phase timers with negligible overhead when disabled
+(wall-clock fallback on CPU). Enable with ``MGPROTO_TIMING=1``; the
+trainer then logs per-phase means every print interval. For kernel-level
+data use rocprofv3 (profiles/).
 """
 
 import os
+import time
 from collections import defaultdict
 from contextlib import contextmanager
 
@@ -16,7 +18,8 @@ class PhaseTimer:
     def __init__(self, enabled=None, device=None):
         if enabled is None:
             enabled = os.environ.get('MGPROTO_TIMING') == '1'
-        self.enabled = enabled and torch.cuda.is_available()
+        self.enabled = enabled
+        self.use_events = enabled and torch.cuda.is_available()
         self._events = []          # (name, start_ev, end_ev)
         self.sums = defaultdict(float)
         self.counts = defaultdict(int)
@@ -25,6 +28,14 @@ class PhaseTimer:
     def phase(self, name):
         if not self.enabled:
             yield
+            return
+        if not self.use_events:            # CPU: wall clock
+            t0 = time.perf_counter()
+            try:
+                yield
+            finally:
+                self.sums[name] += (time.perf_counter() - t0) * 1e3
+                self.counts[name] += 1
             return
         s = torch.cuda.Event(enable_timing=True)
         e = torch.cuda.Event(enable_timing=True)
