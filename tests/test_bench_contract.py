@@ -57,3 +57,18 @@ def test_bench_torchrun_two_ranks():
                        cwd=ROOT)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     _check_contract(_parse_last_json(r.stdout), n_gpus=2)
+
+
+def test_bench_preset_merge():
+    """--preset applies the BASELINE config (arch/classes) while explicit
+    flags still win."""
+    cmd = [sys.executable, 'bench.py', '--preset', 'cars-densenet161',
+           '--steps', '1', '--warmup', '0', '--batch', '2', '--mem', '8',
+           '--mine', '4', '--img', '64']
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=900,
+                       cwd=ROOT)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    rec = _parse_last_json(r.stdout)
+    assert rec['config']['model'] == 'densenet161-mgproto'
+    assert rec['config']['num_classes'] == 196      # preset
+    assert rec['config']['global_batch'] == 2       # explicit flag wins
