@@ -29,7 +29,7 @@ def try_capture(name, fn, warm=3):
         torch.cuda.synchronize()
         print(f'[OK]   {name}')
         return True
-    except Exception:
+    except Exception:  # noqa: BLE001  (any capture failure is the signal)
         torch.cuda.synchronize()
         print(f'[FAIL] {name}')
         traceback.print_exc(limit=8)
@@ -57,7 +57,7 @@ def main():
     try:
         opt = torch.optim.Adam(model.parameters(), lr=1e-4, fused=True,
                                capturable=True)
-    except Exception as e:
+    except Exception as e:  # noqa: BLE001
         print('fused adam failed:', e)
         opt = torch.optim.Adam(model.parameters(), lr=1e-4, capturable=True)
     model.train()
