@@ -90,3 +90,29 @@ def test_eval_purity_csv_driver(mini_cub, tmp_path):
     out = _run('eval_purity.py', mini_cub,
                ('--topK', '2', '--csv', '--log_dir', str(tmp_path / 'logs')))
     assert 'CSV Purity' in out
+
+
+def test_eval_consumes_train_driver_checkpoint(mini_cub, tmp_path):
+    """Full hand-off: a train.py-produced checkpoint (latest.pth with the
+    {'model': ...} wrapper + optimizer state) loads in the eval drivers."""
+    env = dict(os.environ)
+    env['MGPROTO_TINY_TEST'] = '1'
+    cmd = [sys.executable, os.path.join(ROOT, 'train.py'),
+           '-arch', 'resnet18', '-mem_sz', '8', '-mine_level', '3',
+           '-aux_emb_sz', '8', '--epochs', '1',
+           '--out', str(tmp_path / 'run')]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=900,
+                       env=env, cwd=ROOT)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    ckpt = tmp_path / 'run' / 'latest.pth'
+
+    root, _ = mini_cub
+    cmd = [sys.executable, os.path.join(ROOT, 'eval_consistency.py'),
+           '--data_path', str(root), '--resume', str(ckpt),
+           '--base_architecture', 'resnet18', '--prototype_shape', '30',
+           '32', '1', '--nb_classes', '10', '--img_size', '64',
+           '--test_batch_size', '4', '--half_size', '18']
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                       cwd=ROOT)
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    assert 'Consistency Score' in r.stdout
