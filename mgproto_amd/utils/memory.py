@@ -218,7 +218,17 @@ class MemoryBank(nn.Module):
         for c in range(self.num_classes):
             key = prefix + 'cls%d' % c
             if key in state_dict:
-                self.mem[c].copy_(state_dict[key])
+                src = state_dict[key]
+                if src.shape != self.mem[c].shape:
+                    error_msgs.append(
+                        f'size mismatch for {key}: checkpoint has '
+                        f'{tuple(src.shape)}, bank capacity is '
+                        f'{tuple(self.mem[c].shape)} — construct the model '
+                        'with the checkpoint\'s mem_capacity (drivers infer '
+                        'it via infer_ctor_kwargs_from_state)')
+                    state_dict.pop(key)
+                    continue
+                self.mem[c].copy_(src)
                 found += 1
                 state_dict.pop(key)
             elif strict:

@@ -91,3 +91,15 @@ def test_wraparound_after_full():
         bank.push(f, l)
         oracle.push(f, l)
     assert torch.allclose(bank._logical(0), oracle.logical(0))
+
+
+def test_load_capacity_mismatch_reports_cleanly():
+    """A wrong-capacity checkpoint surfaces through load_state_dict's
+    error list (with a pointer to the fix), not a mid-hook crash."""
+    import pytest
+
+    a = MemoryBank(2, 4, capacity=8)    # cap 4/class
+    b = MemoryBank(2, 4, capacity=16)   # cap 8/class
+    sd = a.state_dict()
+    with pytest.raises(RuntimeError, match='mem_capacity'):
+        b.load_state_dict(sd)
