@@ -103,3 +103,37 @@ def test_load_capacity_mismatch_reports_cleanly():
     sd = a.state_dict()
     with pytest.raises(RuntimeError, match='mem_capacity'):
         b.load_state_dict(sd)
+
+
+def test_pull_fix_class_and_dense():
+    bank = MemoryBank(3, 4, capacity=6, mode='all')   # cap 2/class
+    for c in (0, 2):
+        bank.push(torch.full((2, 4), float(c + 1)),
+                  torch.full((2,), c, dtype=torch.long))
+    # fix_class: only flagged classes, logical order
+    data, labels = bank.pull_fix_class(torch.tensor([1, 0, 1]))
+    assert labels.tolist() == [0, 0, 2, 2]
+    assert torch.all(data[:2] == 1.0) and torch.all(data[2:] == 3.0)
+    # empty flag set
+    d2, l2 = bank.pull_fix_class(torch.tensor([0, 1, 0]))
+    assert d2 is None and l2 is None
+    # pull_dense requires full classes; physical order is fine for EM
+    dense = bank.pull_dense(torch.tensor([0, 2]))
+    assert dense.shape == (2, 2, 4)
+    assert torch.all(dense[0] == 1.0) and torch.all(dense[1] == 3.0)
+    assert bank.full_mask().tolist() == [True, False, True]
+
+
+def test_pull_fix_length_shapes():
+    torch.manual_seed(0)
+    bank = MemoryBank(3, 4, capacity=6, fix_length_mult=2)  # pull_num = 4
+    for c in range(3):
+        bank.push(torch.randn(2, 4), torch.full((2,), c, dtype=torch.long))
+    onehot = torch.tensor([[1., 0., 1.], [0., 1., 0.]])
+    out, lab = bank.pull_fix_length(onehot)
+    assert lab is None
+    assert out.shape == (4, 2, 4)    # [pull_num, B, d]
+    # a row flagging a class with no data -> None
+    bank2 = MemoryBank(3, 4, capacity=6)
+    bank2.push(torch.randn(2, 4), torch.zeros(2, dtype=torch.long))
+    assert bank2.pull_fix_length(torch.tensor([[0., 1., 0.]]))[0] is None

@@ -170,3 +170,31 @@ def test_training_without_aux_criterion():
                          coefs={'crs_ent': 1, 'mine': 0.2, 'aux': 0.5},
                          log=lambda *a: None, amp_dtype='off', print_every=0)
     assert res['aux_loss'] == 0.0
+
+
+def test_rf_prototypes_plural_and_helpers():
+    from mgproto_amd.utils.receptive_field import (
+        compute_proto_layer_rf_info_v2, compute_rf_prototype,
+        compute_rf_prototypes)
+    from mgproto_amd.utils.helpers import (list_of_distances, make_one_hot,
+                                           find_high_activation_crop)
+    import numpy as np
+
+    info = compute_proto_layer_rf_info_v2(
+        224, [7, 3, 3, 3], [2, 2, 2, 2], [3, 1, 1, 1], 1)
+    singles = [compute_rf_prototype(224, [i, 3, 5], info) for i in range(2)]
+    plural = compute_rf_prototypes(224, [[0, 3, 5], [1, 3, 5]], info)
+    assert [list(p) for p in plural] == [list(s) for s in singles]
+
+    # list_of_distances == pairwise squared L2
+    X, Y = torch.randn(4, 3), torch.randn(5, 3)
+    D = list_of_distances(X, Y)
+    assert torch.allclose(D, torch.cdist(X, Y).pow(2), atol=1e-5)
+
+    oh = torch.zeros(3, 4)
+    make_one_hot(torch.tensor([1, 0, 3]), oh)
+    assert oh.argmax(1).tolist() == [1, 0, 3] and oh.sum() == 3
+
+    act = np.zeros((10, 10)); act[4:6, 6:8] = 5.0
+    y1, y2, x1, x2 = find_high_activation_crop(act, percentile=95)
+    assert (y1, y2, x1, x2) == (4, 6, 6, 8)
