@@ -195,6 +195,8 @@ def test_rf_prototypes_plural_and_helpers():
     make_one_hot(torch.tensor([1, 0, 3]), oh)
     assert oh.argmax(1).tolist() == [1, 0, 3] and oh.sum() == 3
 
-    act = np.zeros((10, 10)); act[4:6, 6:8] = 5.0
+    # >5% of pixels must exceed the 95th percentile for a tight crop
+    # (reference helpers.py:38 thresholds on the percentile of ALL pixels)
+    act = np.zeros((10, 10)); act[4:7, 6:8] = 5.0
     y1, y2, x1, x2 = find_high_activation_crop(act, percentile=95)
-    assert (y1, y2, x1, x2) == (4, 6, 6, 8)
+    assert (y1, y2, x1, x2) == (4, 7, 6, 8)
