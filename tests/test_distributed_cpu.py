@@ -415,3 +415,31 @@ def impl_ood_eval_single():
                                    log=lambda *a: None, amp_dtype='off')
     return {'acc': float(results['acc']),
             'FPR95_1': float(results['FPR95_1'])}
+
+
+def test_comm_gather_primitives():
+    res = _run_workers(impl_comm_gather_primitives, world=2)
+    assert res[0] == res[1] == 'ok'
+
+
+def impl_comm_gather_primitives(rank, world):
+    from mgproto_amd.parallel import Comm
+
+    comm = Comm(backend='gloo')
+    # fixed: [world, *shape], rank-major
+    t = torch.full((2, 3), float(rank))
+    g = comm.all_gather_fixed(t)
+    assert g.shape == (2, 2, 3)
+    assert torch.all(g[0] == 0.0) and torch.all(g[1] == 1.0)
+    # scalar input gets a leading dim
+    s = comm.all_gather_fixed(torch.tensor(float(rank)))
+    assert s.flatten().tolist() == [0.0, 1.0]
+    # varlen concatenates in rank order
+    v = comm.all_gather_varlen(torch.arange(2 + rank).float())
+    assert v.tolist() == [0.0, 1.0, 0.0, 1.0, 2.0]
+    # async reduce completes
+    x = torch.ones(4)
+    w = comm.all_reduce_sum_async(x)
+    w.wait()
+    assert torch.all(x == world)
+    return 'ok'
