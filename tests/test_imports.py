@@ -27,3 +27,17 @@ def test_top_level_drivers_import():
     import eval_purity        # noqa: F401
     import eval_stability     # noqa: F401
     import __graft_entry__    # noqa: F401
+
+
+def test_check_env_tool_runs():
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run([sys.executable, 'tools/check_env.py'],
+                       capture_output=True, text=True, timeout=120, cwd=root)
+    # exit 1 here (no GPU in CI) but every non-GPU check must PASS
+    assert 'torch' in r.stdout
+    for line in r.stdout.splitlines():
+        if line.startswith('FAIL'):
+            assert 'GPU' in line or 'gfx950' in line or 'loads' in line, line
