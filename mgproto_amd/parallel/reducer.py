@@ -92,6 +92,12 @@ class BucketedGradReducer:
         bucket = self.buckets[bi]
         flat = torch._utils._flatten_dense_tensors(
             [p.grad for p in bucket])
+        # single-tensor buckets: _flatten returns a VIEW of p.grad — reduce
+        # a copy instead, so p.grad keeps its local value until finalize()
+        # (an async in-place reduce would make grads read between backward
+        # and finalize, e.g. for clipping, undefined)
+        if len(bucket) == 1 and flat.data_ptr() == bucket[0].grad.data_ptr():
+            flat = flat.clone()
         flat.div_(self.comm.world_size)
         work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, async_op=True)
         self._flat[bi] = flat

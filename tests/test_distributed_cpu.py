@@ -443,3 +443,47 @@ def impl_comm_gather_primitives(rank, world):
     w.wait()
     assert torch.all(x == world)
     return 'ok'
+
+
+def test_grad_reducer_random_configs():
+    """Reducer over randomized layer shapes and bucket sizes (1 bucket ..
+    one-per-param) against manually averaged gradients, several steps."""
+    res = _run_workers(impl_grad_reducer_random_configs, world=2)
+    assert res[0] == res[1] == 'ok'
+
+
+def impl_grad_reducer_random_configs(rank, world):
+    import torch.distributed as dist
+
+    from mgproto_amd.parallel import Comm, BucketedGradReducer
+
+    comm = Comm(backend='gloo')
+    for trial in range(4):
+        g = torch.Generator().manual_seed(900 + trial)   # same on all ranks
+        sizes = [int(torch.randint(1, 2000, (1,), generator=g)) for _ in range(6)]
+        layers = []
+        d_in = 8
+        for s in sizes:
+            layers.append(torch.nn.Linear(d_in, s))
+            d_in = s
+        model = torch.nn.Sequential(*layers)
+        comm.broadcast_module(model)
+        bucket_mb = [1, 1000][trial % 2]   # many tiny buckets vs one big
+        red = BucketedGradReducer(model, comm, bucket_mb=bucket_mb)
+        for step in range(2):
+            x = torch.randn(3, 8, generator=torch.Generator().manual_seed(
+                rank * 17 + step + trial))        # different data per rank
+            red.prepare()
+            model.zero_grad(set_to_none=True)
+            model(x).square().mean().backward()
+            # manual expectation BEFORE finalize touches grads
+            expected = [p.grad.clone() for p in model.parameters()]
+            for e in expected:
+                dist.all_reduce(e)
+                e.div_(world)
+            red.finalize()
+            for p, e in zip(model.parameters(), expected):
+                assert torch.allclose(p.grad, e, atol=1e-6), \
+                    (trial, step, bucket_mb)
+        red.remove()
+    return 'ok'
