@@ -442,6 +442,13 @@ def impl_comm_gather_primitives(rank, world):
     w = comm.all_reduce_sum_async(x)
     w.wait()
     assert torch.all(x == world)
+    # broadcast_buffers: rank 0's float buffers win; ints untouched
+    m = torch.nn.BatchNorm2d(3)
+    m.running_mean.fill_(float(rank))
+    m.num_batches_tracked.fill_(rank)
+    comm.broadcast_buffers(m)
+    assert torch.all(m.running_mean == 0.0)
+    assert int(m.num_batches_tracked) == rank   # non-float: left alone
     return 'ok'
 
 
