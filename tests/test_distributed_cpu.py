@@ -494,3 +494,54 @@ def impl_grad_reducer_random_configs(rank, world):
                     (trial, step, bucket_mb)
         red.remove()
     return 'ok'
+
+
+def test_distributed_enqueue_random_streams():
+    """Randomized multi-step enqueue streams on 2 ranks: banks stay
+    bit-identical AND equal a single-process push of the rank-ordered
+    concatenation of every step's candidates."""
+    res = _run_workers(impl_distributed_enqueue_random_streams, world=2)
+    assert res[0] == res[1] == 'ok'
+
+
+def impl_distributed_enqueue_random_streams(rank, world):
+    from mgproto_amd.model import construct_MGProto
+    from mgproto_amd.parallel import Comm
+    from mgproto_amd.parallel.state_sync import DistributedEnqueue
+    from mgproto_amd.utils.memory import MemoryBank
+
+    comm = Comm(backend='gloo')
+    C, K, d, cap = 5, 2, 8, 4
+    torch.manual_seed(0)
+    model = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                              prototype_shape=(C * K, d, 1, 1),
+                              num_classes=C, add_on_layers_type='regular',
+                              sz_embedding=8, mem_capacity=cap, mine_K=2)
+    enq = DistributedEnqueue(model, comm, max_items=6)
+    shadow = MemoryBank(C, d, capacity=C * cap)   # single-process oracle
+
+    for step in range(12):
+        per_rank = []
+        for r in range(world):
+            g = torch.Generator().manual_seed(step * 100 + r)
+            M = int(torch.randint(0, 7, (1,), generator=g))
+            feats = torch.randn(M, d, generator=g)
+            labels = torch.randint(0, C, (M,), generator=g)
+            per_rank.append((feats, labels))
+        feats, labels = per_rank[rank]
+        enq(feats, labels)
+        # oracle: rank-ordered concatenation with sentinel padding, exactly
+        # as the fixed-size all-gather delivers it
+        pf, pl = [], []
+        for f, l in per_rank:
+            pad = 6 - f.shape[0]
+            pf.append(torch.cat([f, torch.zeros(pad, d)]))
+            pl.append(torch.cat([l, torch.full((pad,), C, dtype=torch.long)]))
+        shadow.push(torch.cat(pf), torch.cat(pl))
+
+    assert torch.equal(model.queue.mem, shadow.mem)
+    assert torch.equal(model.queue.mem_len, shadow.mem_len)
+    # cross-rank bit-identity
+    other = comm.all_gather_fixed(model.queue.mem.reshape(1, -1))
+    assert torch.equal(other[0], other[1])
+    return 'ok'
