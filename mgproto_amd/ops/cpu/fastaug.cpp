@@ -103,6 +103,9 @@ torch::Tensor warp_jitter_normalize(torch::Tensor img_u8, torch::Tensor coeffs,
     img_u8 = img_u8.contiguous();
     coeffs = coeffs.to(torch::kFloat64).contiguous();
     TORCH_CHECK(coeffs.numel() == 8, "coeffs must have 8 elements");
+    TORCH_CHECK(out_size > 0, "out_size must be positive");
+    TORCH_CHECK(mean.numel() == 3 && stdv.numel() == 3,
+                "mean/std must have 3 elements");
     const int Hh = img_u8.size(0), Ww = img_u8.size(1);
     const int S = (int)out_size;
     const double* cf = coeffs.data_ptr<double>();
