@@ -179,14 +179,19 @@ def test_mining_grad_flows_to_backbone(small_model):
     assert m.prototype_means.grad is None or m.prototype_means.grad.abs().sum() == 0
 
 
-def test_forward_matches_manual_pipeline():
+@pytest.mark.parametrize('C,K,d,T', [
+    (5, 3, 16, 4),    # baseline shape
+    (2, 1, 8, 1),     # minimal: one prototype/class, top-1
+    (3, 4, 24, 12),   # T close to HW corner (HW=16 at img 64)
+    (7, 2, 32, 3),    # odd class count
+])
+def test_forward_matches_manual_pipeline(C, K, d, T):
     """Full-forward differential oracle: model.forward(x, None) must equal
     the manually composed pipeline (direct log-prob formula -> exp -> top-T
     -> pi-weighted mixture -> log), i.e. the reference's semantics
     (model.py:208-254) assembled from first principles."""
     from mgproto_amd.ops import reference as R
     torch.manual_seed(0)
-    C, K, d, T = 5, 3, 16, 4
     m = construct_MGProto('resnet18', pretrained=False, img_size=64,
                           prototype_shape=(C * K, d, 1, 1), num_classes=C,
                           add_on_layers_type='regular', sz_embedding=8,
