@@ -286,7 +286,8 @@ def em_m_step_grads(x: torch.Tensor, log_resp: torch.Tensor, wlp: torch.Tensor,
     w = torch.exp(-dist)
     eye = torch.eye(K, device=x.device, dtype=torch.bool)
     w = w.masked_fill(eye, 0.0)
-    S = K * K - K
+    # K=1: no pairs to repel — the reference's 0/0 here would NaN the means
+    S = max(K * K - K, 1)
     grad_div = -(4.0 / S) * torch.einsum('gik,gikd->gid', w, diff)     # [G, K, d]
 
     return grad_nll + lamda * grad_div, pi_unnorm

@@ -309,3 +309,24 @@ def test_update_gmm_matches_per_class_loop():
 
     # dirty flags consumed
     assert not model.memory_updated_cls.any()
+
+
+def test_update_gmm_single_component_class():
+    """K=1 per class: the diversity term has no pairs — EM must stay
+    finite (the reference's 0/0 diversity division would NaN the means)."""
+    C, K, d, cap = 3, 1, 8, 4
+    model = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                              prototype_shape=(C * K, d, 1, 1),
+                              num_classes=C, add_on_layers_type='regular',
+                              sz_embedding=8, mem_capacity=cap, mine_K=1)
+    g = torch.Generator().manual_seed(0)
+    for c in range(C):
+        model.queue.push(F.normalize(torch.randn(cap, d, generator=g), dim=1),
+                         torch.full((cap,), c, dtype=torch.long))
+        model.memory_updated_cls[c] = True
+    before = model.prototype_means.data.clone()
+    model.update_GMM()
+    after = model.prototype_means.data
+    assert torch.isfinite(after).all()
+    assert not torch.equal(after, before)          # NLL grad still moves them
+    assert torch.isfinite(model.last_layer.weight).all()
