@@ -265,7 +265,10 @@ class MGProto(nn.Module):
 
         probs = ops.gmm_scores(feat, self.prototype_means, self.prototype_covs,
                                apply_exp=True)                     # [N, P]
-        vals, idx = ops.topk_hw(probs.view(B, HW, P), self.mine_T)  # [B, P, T]
+        # mining pool cannot exceed the latent grid (e.g. mine_T=20 on a
+        # small-image 4x4 grid — the reference's topk would error out)
+        vals, idx = ops.topk_hw(probs.view(B, HW, P),
+                                min(self.mine_T, HW))       # [B, P, T]
 
         if gt is not None:
             vals = ops.mask_wrong_class(vals, gt, self.prototype_class_identity)

@@ -330,3 +330,17 @@ def test_update_gmm_single_component_class():
     assert torch.isfinite(after).all()
     assert not torch.equal(after, before)          # NLL grad still moves them
     assert torch.isfinite(model.last_layer.weight).all()
+
+
+def test_mine_pool_larger_than_grid():
+    """mine_T > HW clamps to the grid size instead of erroring (the
+    reference's topk(T) would throw on small latent grids)."""
+    C, K, d = 3, 2, 8
+    model = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                              prototype_shape=(C * K, d, 1, 1),
+                              num_classes=C, add_on_layers_type='regular',
+                              sz_embedding=8, mem_capacity=4, mine_K=64)
+    x = torch.randn(2, 3, 64, 64)
+    out, _ = model(x, torch.tensor([0, 1]))
+    assert out.shape == (2, C, 16)      # 4x4 grid -> 16 mining levels
+    assert torch.isfinite(out).all()
