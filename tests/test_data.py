@@ -237,3 +237,22 @@ def test_scaled_jpeg_decode(tmp_path):
     ds2 = ImageFolder(str(tmp_path), None, decode_size=448)
     p2 = [p for p, _ in ds2.samples if p.endswith('.png')][0]
     assert ds2.loader(p2).size == (1200, 900)
+
+
+def test_device_prefetcher_cpu_passthrough():
+    """CPU fallback: yields every batch unchanged, in order; and a
+    gpu-marked variant checks the stream path separately."""
+    from torch.utils.data import DataLoader
+
+    from mgproto_amd.data.prefetch import DevicePrefetcher
+    from mgproto_amd.data.synthetic import SyntheticImages
+
+    ds = SyntheticImages(n=10, num_classes=3, img_size=32)
+    loader = DataLoader(ds, batch_size=4)
+    pf = DevicePrefetcher(loader, torch.device('cpu'))
+    assert len(pf) == len(loader)
+    batches = list(pf)
+    ref = list(loader)
+    assert len(batches) == len(ref) == 3
+    for (a, b, c), (x, y, z) in zip(batches, ref):
+        assert torch.equal(a, x) and torch.equal(b, y) and torch.equal(c, z)

@@ -74,3 +74,24 @@ def test_gpu_matches_cpu_model_outputs():
     # fp32 end to end on both paths; conv nondeterminism tolerance
     assert torch.allclose(out_gpu.cpu(), out_cpu, atol=1e-3, rtol=1e-3)
     assert torch.allclose(emb_gpu.cpu(), emb_cpu, atol=1e-4, rtol=1e-4)
+
+
+@pytest.mark.gpu
+def test_device_prefetcher_stream_path():
+    """Prefetcher on CUDA: same batches, device tensors, overlap-safe."""
+    from torch.utils.data import DataLoader
+
+    from mgproto_amd.data.prefetch import DevicePrefetcher
+    from mgproto_amd.data.synthetic import SyntheticImages
+
+    ds = SyntheticImages(n=12, num_classes=3, img_size=32)
+    loader = DataLoader(ds, batch_size=4, pin_memory=True)
+    pf = DevicePrefetcher(loader, torch.device('cuda', 0))
+    got = list(pf)
+    ref = list(DataLoader(ds, batch_size=4))
+    assert len(got) == 3
+    for (a, b, c), (x, y, z) in zip(got, ref):
+        assert a.is_cuda and b.is_cuda
+        assert torch.equal(a.cpu(), x) and torch.equal(b.cpu(), y)
+        # consume with a compute op to exercise the event ordering
+        _ = (a * 2).sum().item()

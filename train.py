@@ -86,6 +86,10 @@ def main():
     parser.add_argument('--resume', type=str, default=None)
     parser.add_argument('--out', type=str, default=None)
     parser.add_argument('--ood-eval', action='store_true')
+    parser.add_argument('--prefetch', action='store_true',
+                        help='stage next-batch H2D copies on a side stream '
+                             '(real-data runs; the synthetic bench path is '
+                             'already device-resident)')
     parser.add_argument('--addon', type=str, default=None)
     parser.add_argument('--no-auto-resume', action='store_true',
                         help='do not resume from an existing latest.pth / '
@@ -149,6 +153,9 @@ def main():
 
     loaders = build_loaders(cfg, comm)
     train_loader, push_loader, test_loader, ood1_loader, ood2_loader = loaders
+    if args.prefetch and device.type == 'cuda':
+        from mgproto_amd.data.prefetch import DevicePrefetcher
+        train_loader = DevicePrefetcher(train_loader, device)
 
     ppnet = construct_MGProto(
         base_architecture=args.arch, pretrained=True, img_size=cfg.img_size,
