@@ -3,9 +3,12 @@
 Re-implementation of the reference backbone contract
 (``/root/reference/models/vgg_features.py``): VGG-11/13/16/19 (+BN variants)
 trunks with the classifier removed, options to drop the final max-pool
-(default: dropped, so output stride is 16) and the final ReLU, and per-layer
-``conv_info()``. Module naming matches torchvision (``features.N``) so
-reference checkpoints load directly.
+(default: dropped, so output stride is 16) and the final ReLU
+(reference vgg_features.py:58-94), per-layer ``conv_info()``
+(vgg_features.py:42-56), and ``VGG_vanilla`` (vgg_features.py:110-124).
+Module naming matches torchvision (``features.N``) so reference
+checkpoints load directly. The norm->relu pairs run as
+FusedBatchNorm2d(fused_relu=True) on GPU (models/fused_bn.py).
 """
 
 import os
