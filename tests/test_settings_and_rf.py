@@ -200,3 +200,22 @@ def test_rf_prototypes_plural_and_helpers():
     act = np.zeros((10, 10)); act[4:7, 6:8] = 5.0
     y1, y2, x1, x2 = find_high_activation_crop(act, percentile=95)
     assert (y1, y2, x1, x2) == (4, 7, 6, 8)
+
+
+@pytest.mark.parametrize('arch', ['resnet18', 'resnet50', 'densenet121',
+                                  'vgg11', 'vgg16_bn'])
+def test_conv_info_predicts_actual_grid(arch):
+    """conv_info() must describe the real forward: the RF recurrence's
+    output size equals the actual feature-map size (the reference counts
+    a skipped max-pool, reference resnet_features.py:199)."""
+    import torch
+
+    from mgproto_amd.model import base_architecture_to_features
+    from mgproto_amd.utils.receptive_field import compute_proto_layer_rf_info_v2
+
+    f = base_architecture_to_features[arch](pretrained=False)
+    ks, ss, ps = f.conv_info()
+    info = compute_proto_layer_rf_info_v2(96, ks, ss, ps, 1)
+    with torch.no_grad():
+        out = f(torch.randn(1, 3, 96, 96))
+    assert info[0] == out.shape[-1], (info[0], out.shape)
