@@ -147,3 +147,20 @@ def test_enqueue_and_em_on_gpu():
     wlp_c, lr_c = R.em_e_step(x.cpu(), means.cpu(), covs.cpu(), pi.cpu())
     assert torch.allclose(wlp.cpu(), wlp_c, atol=1e-4)
     assert torch.allclose(lr_.cpu(), lr_c, atol=1e-4)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize('d', [144, 60])   # beyond the kernel envelope
+def test_gmm_dispatch_fallback_out_of_envelope(d):
+    """d > 128 or d % 8 != 0 routes to the GEMM fallback (hipBLASLt via
+    addmm) transparently — same numerics, gradient flows."""
+    import mgproto_amd.ops as O
+    dev = torch.device('cuda', 0)
+    feat, means, covs = make_gmm(256, 40, d, dev, seed=9)
+    feat = feat.clone().requires_grad_(True)
+    out = O.gmm_scores(feat, means, covs, apply_exp=True)
+    ref = R.gmm_logprob_direct(feat.detach().float(),
+                               means.float(), covs.float()).exp()
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+    out.sum().backward()
+    assert feat.grad is not None and torch.isfinite(feat.grad).all()
