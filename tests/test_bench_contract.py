@@ -72,3 +72,14 @@ def test_bench_preset_merge():
     assert rec['config']['model'] == 'densenet161-mgproto'
     assert rec['config']['num_classes'] == 196      # preset
     assert rec['config']['global_batch'] == 2       # explicit flag wins
+
+
+def test_bench_flags_reflected_in_config():
+    """--no-em / --eager toggle the measured step and are reported."""
+    cmd = [sys.executable, 'bench.py', *TINY, '--no-em', '--eager']
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                       cwd=ROOT)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    rec = _parse_last_json(r.stdout)
+    assert rec['config']['em_active'] is False
+    assert rec['config']['hip_graph'] is False
