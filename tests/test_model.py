@@ -344,3 +344,32 @@ def test_mine_pool_larger_than_grid():
     out, _ = model(x, torch.tensor([0, 1]))
     assert out.shape == (2, C, 16)      # 4x4 grid -> 16 mining levels
     assert torch.isfinite(out).all()
+
+
+def test_gradient_flow_partition():
+    """Exactly the gradient-trained parameter sets get grads: backbone,
+    add-on, embedding. Prototype means/covs and the mixture head are
+    EM-state (reference model.py:64,264-265) and must stay grad-free."""
+    torch.manual_seed(0)
+    C, K, d = 3, 2, 8
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(C * K, d, 1, 1), num_classes=C,
+                          add_on_layers_type='regular', sz_embedding=8,
+                          mem_capacity=4, mine_K=2)
+    m.train()
+    x = torch.randn(2, 3, 64, 64)
+    y = torch.tensor([0, 1])
+    out, aux = m(x, y)
+    (F.cross_entropy(out[:, :, 0], y) + aux.square().mean()).backward()
+
+    def has_grad(params):
+        return any(p.grad is not None and p.grad.abs().sum() > 0
+                   for p in params)
+
+    assert has_grad(m.features.parameters())
+    assert has_grad(m.add_on_layers.parameters())
+    assert has_grad(m.embedding.parameters())
+    assert m.prototype_means.grad is None
+    assert m.prototype_covs.grad is None
+    assert m.last_layer.weight.grad is None
+    assert m.iteration_counter.grad is None
