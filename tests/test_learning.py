@@ -67,7 +67,7 @@ def test_training_learns_separable_data():
             loss.backward()
             opt.step()
             model.update_GMM()  # EM active every step (update_interval=1)
-            losses.append(float(loss))
+            losses.append(float(loss.detach()))
 
     # loss must drop substantially and accuracy must beat chance by a lot
     assert losses[-1] < 0.6 * losses[0], (losses[0], losses[-1])
