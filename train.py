@@ -142,6 +142,16 @@ def main():
         torch.backends.cudnn.benchmark = True
 
     model_dir = args.out or 'saved_models/{}/{}/'.format(args.arch, datestr())
+    if comm is not None and comm.is_distributed:
+        # ranks must agree on the run dir (datestr() can straddle a minute
+        # boundary): rank 0's choice wins
+        buf = torch.zeros(512, dtype=torch.uint8, device=comm.device)
+        if rank == 0:
+            raw = model_dir.encode()[:512]
+            buf[:len(raw)] = torch.tensor(list(raw), dtype=torch.uint8,
+                                          device=comm.device)
+        comm.broadcast(buf)
+        model_dir = bytes(b for b in buf.cpu().tolist() if b).decode()
     if rank == 0:
         makedir(model_dir)
         makedir(os.path.join(model_dir, 'img'))
