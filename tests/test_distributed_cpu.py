@@ -545,3 +545,41 @@ def impl_distributed_enqueue_random_streams(rank, world):
     other = comm.all_gather_fixed(model.queue.mem.reshape(1, -1))
     assert torch.equal(other[0], other[1])
     return 'ok'
+
+
+def test_distributed_push_empty_rank_shard():
+    """More ranks than images: a rank with an EMPTY push shard must not
+    deadlock or diverge — its varlen gathers contribute zero candidates."""
+    _run_workers(impl_distributed_push_empty_shard, world=3)
+
+
+def impl_distributed_push_empty_shard(rank, world):
+    from torch.utils.data import DataLoader, Subset
+
+    from mgproto_amd.data import SyntheticImages
+    from mgproto_amd.engine import push_prototypes
+    from mgproto_amd.parallel import Comm
+
+    comm = Comm(backend='gloo')
+    model = _tiny_model()
+    model.eval()
+    ds = SyntheticImages(n=2, num_classes=5, img_size=64, normalize=False)
+
+    def collate(batch):
+        return (torch.stack([b[0] for b in batch]),
+                torch.tensor([b[1] for b in batch]),
+                torch.tensor([b[2] for b in batch]))
+
+    shard = Subset(ds, list(range(rank, len(ds), world)))   # rank 2: empty
+    loader = DataLoader(shard, batch_size=2, collate_fn=collate)
+    chosen = push_prototypes(loader, model, log=lambda *a: None, comm=comm)
+
+    model2 = _tiny_model()
+    model2.eval()
+    chosen_single = push_prototypes(DataLoader(ds, batch_size=2,
+                                               collate_fn=collate),
+                                    model2, log=lambda *a: None)
+    assert chosen == chosen_single
+    assert torch.allclose(model.prototype_means.data,
+                          model2.prototype_means.data, atol=1e-5)
+    return 'ok'
