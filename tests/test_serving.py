@@ -126,3 +126,20 @@ def test_serve_cli_loads_any_config_checkpoint(tmp_path):
                        text=True, timeout=300, cwd=root)
     assert r.returncode == 0 and 'LOADED-OK' in r.stdout, \
         r.stdout[-2000:] + r.stderr[-2000:]
+
+
+def test_serve_bench_tool_cpu(tmp_path):
+    import json
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, 'tools/serve_bench.py', '--arch', 'resnet18',
+         '--classes', '4', '--proto-per-class', '2', '--proto-dim', '16',
+         '--img', '64', '--batch', '2', '--iters', '3', '--warmup', '1'],
+        capture_output=True, text=True, timeout=600, cwd=root)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    rec = json.loads([l for l in r.stdout.splitlines()
+                      if l.startswith('{')][-1])
+    assert rec['mode'] == 'eager' and rec['images_per_sec'] > 0

@@ -19,6 +19,9 @@ timeout 600 python bench.py --steps 30 --warmup 10 > gpurun_out/r2/bench_default
 timeout 600 bash -c 'MGPROTO_BN_MASK=1 python bench.py --steps 30 --warmup 10' > gpurun_out/r2/bench_bnmask.json 2>/dev/null
 timeout 600 python bench.py --steps 30 --warmup 10 --no-graph > gpurun_out/r2/bench_nograph.json 2>/dev/null
 
+# 3b. serving latency (eager vs captured forward)
+timeout 600 python tools/serve_bench.py --batch 8 --iters 50 > gpurun_out/r2/serve_bench.jsonl 2>/dev/null
+
 # 4. kernel-level profile of the default config (small CSVs only)
 cd /tmp && export TMPDIR=/tmp && cd - >/dev/null
 timeout 900 rocprofv3 --kernel-trace --stats --output-format csv -d /tmp/prof_r2 \
