@@ -110,3 +110,24 @@ def test_rng_resume(tmp_path):
     st = load_train_state(str(tmp_path / 'st.pt'), m)
     got = (random.random(), float(np.random.random()), float(torch.rand(1)))
     assert got == expect and st['epoch'] == 3
+
+
+def test_inspect_checkpoint_tool(tmp_path):
+    import os
+    import subprocess
+    import sys
+
+    from mgproto_amd.model import construct_MGProto
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(6, 16, 1, 1), num_classes=3,
+                          add_on_layers_type='regular', sz_embedding=8,
+                          mem_capacity=4, mine_K=2)
+    p = tmp_path / 'm.pth'
+    torch.save(m.state_dict(), str(p))
+    r = subprocess.run([sys.executable, 'tools/inspect_checkpoint.py',
+                        str(p)], capture_output=True, text=True,
+                       timeout=300, cwd=root)
+    assert r.returncode == 0, r.stderr
+    assert 'C=3 classes x K=2' in r.stdout
+    assert 'memory bank: 3 classes x cap 4' in r.stdout
