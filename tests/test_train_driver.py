@@ -121,3 +121,12 @@ def test_train_driver_ood_eval(tmp_path):
     _run_train(tmp_path, extra=('--ood-eval',))
     log_text = open(tmp_path / 'run' / 'train.log').read()
     assert 'FPR95_1' in log_text and 'FPR95_2' in log_text
+
+
+def test_summarize_metrics_tool(tmp_path):
+    _run_train(tmp_path)
+    r = subprocess.run([sys.executable, 'tools/summarize_metrics.py',
+                        str(tmp_path / 'run' / 'metrics.jsonl')],
+                       capture_output=True, text=True, timeout=120, cwd=ROOT)
+    assert r.returncode == 0, r.stderr
+    assert 'best test acc' in r.stdout and 'epoch' in r.stdout
