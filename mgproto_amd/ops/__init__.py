@@ -16,8 +16,7 @@ import torch
 
 from . import reference
 from .reference import (gmm_expand_params, gmm_logprob_direct, mask_wrong_class,  # noqa: F401
-                        mixture_head, gather_patch_features, enqueue_candidates,
-                        em_e_step, em_m_step_grads)
+                        mixture_head, gather_patch_features, enqueue_candidates)
 
 _EXT = None
 _EXT_ERR: Optional[str] = None
@@ -150,6 +149,51 @@ class _TopkHW(torch.autograd.Function):
 def topk_hw(probs: torch.Tensor, T: int) -> Tuple[torch.Tensor, torch.Tensor]:
     """[B, HW, P] -> (values [B, P, T] desc-sorted, indices [B, P, T])."""
     return _TopkHW.apply(probs, T)
+
+
+def _em_hip_usable(x: torch.Tensor, K: int) -> bool:
+    """K6/K7 HIP kernels: opt-in (MGPROTO_HIP_EM=1) until GPU-validated;
+    the default EM path is the rocBLAS baddbmm form (reference.py)."""
+    return (x.is_cuda and os.environ.get('MGPROTO_HIP_EM') == '1'
+            and x.shape[-1] <= 128 and K <= 32
+            and _load_extension() is not None)
+
+
+def em_e_step(x: torch.Tensor, means: torch.Tensor, covs: torch.Tensor,
+              pi: torch.Tensor, eps: float = 1e-10):
+    """Batched EM E-step. Same semantics as the oracle
+    (reference.py em_e_step == reference model.py:303-336)."""
+    if not _em_hip_usable(x, means.shape[1]):
+        return reference.em_e_step(x, means, covs, pi, eps=eps)
+    ext = _load_extension()
+    d = x.shape[-1]
+    sig = (covs + eps).float()
+    inv_var = 1.0 / (sig * sig)
+    A = (means.float() * inv_var).contiguous()                   # [G, K, d]
+    B = (-0.5 * inv_var).contiguous()
+    bias = (-0.5 * d * reference.LOG_2PI
+            - torch.log(sig).sum(-1)
+            - 0.5 * (means.float() * means.float() * inv_var).sum(-1)
+            + torch.log(pi.float() + eps)).contiguous()          # [G, K]
+    wlp, logresp = ext.em_estep(x.float().contiguous(), A, B, bias)
+    return wlp, logresp
+
+
+def em_m_step_grads(x: torch.Tensor, log_resp: torch.Tensor,
+                    wlp: torch.Tensor, means: torch.Tensor,
+                    covs: torch.Tensor, alpha: float = 0.1,
+                    lamda: float = 1.0, eps: float = 1e-10):
+    """Batched EM M-step closed-form grads."""
+    if not _em_hip_usable(x, means.shape[1]):
+        return reference.em_m_step_grads(x, log_resp, wlp, means, covs,
+                                         alpha=alpha, lamda=lamda, eps=eps)
+    ext = _load_extension()
+    grad, pi_unnorm = ext.em_mstep(x.float().contiguous(),
+                                   log_resp.float().contiguous(),
+                                   means.float().contiguous(),
+                                   covs.float().contiguous(),
+                                   alpha, lamda, eps)
+    return grad, pi_unnorm
 
 
 def argmax_hw(probs: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:

@@ -544,6 +544,13 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
                                   bool training, bool relu, bool has_res,
                                   bool use_mask);
 
+// batched EM kernels (em_kernels.hip)
+std::vector<torch::Tensor> em_estep(torch::Tensor x, torch::Tensor A,
+                                    torch::Tensor B, torch::Tensor bias);
+std::vector<torch::Tensor> em_mstep(torch::Tensor x, torch::Tensor logresp,
+                                    torch::Tensor means, torch::Tensor covs,
+                                    double alpha, double lamda, double eps);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gmm_fwd", &gmm_fwd, "fused GMM score GEMM forward (MFMA f32)");
     m.def("gmm_bwd", &gmm_bwd, "GMM score GEMM backward wrt features");
@@ -551,4 +558,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("argmax_hw", &argmax_hw, "per-(b,p) argmax over HW");
     m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+Add)(+ReLU) forward, NHWC bf16");
     m.def("bn_bwd", &bn_bwd, "fused BatchNorm(+Add)(+ReLU) backward, NHWC bf16");
+    m.def("em_estep", &em_estep, "batched EM e-step (wlp + log-resp)");
+    m.def("em_mstep", &em_mstep, "batched EM m-step (closed-form grads + pi)");
 }
