@@ -14,6 +14,12 @@ timeout 600 bash -c 'MGPROTO_BN_MASK=1 python -m pytest tests/test_fused_bn_gpu.
     > gpurun_out/r2/bn_mask_tests.log 2>&1
 echo "bn-mask tests: $?" >> gpurun_out/r2/summary.txt
 
+# 2b. HIP EM kernels: parity tests + A/B bench
+timeout 600 bash -c 'MGPROTO_HIP_EM=1 python -m pytest tests/test_em_hip_gpu.py -x -q' \
+    > gpurun_out/r2/em_hip_tests.log 2>&1
+echo "em-hip tests: $?" >> gpurun_out/r2/summary.txt
+timeout 600 bash -c 'MGPROTO_HIP_EM=1 python bench.py --steps 30 --warmup 10' > gpurun_out/r2/bench_emhip.json 2>/dev/null
+
 # 3. flagship bench: default, BN-mask on, eager (for the graph delta)
 timeout 600 python bench.py --steps 30 --warmup 10 > gpurun_out/r2/bench_default.json 2>gpurun_out/r2/bench_default.log
 timeout 600 bash -c 'MGPROTO_BN_MASK=1 python bench.py --steps 30 --warmup 10' > gpurun_out/r2/bench_bnmask.json 2>/dev/null
