@@ -16,7 +16,7 @@ import torch
 
 from . import reference
 from .reference import (gmm_expand_params, gmm_logprob_direct, mask_wrong_class,  # noqa: F401
-                        mixture_head, gather_patch_features, enqueue_candidates)
+                        mixture_head, gather_patch_features)
 
 _EXT = None
 _EXT_ERR: Optional[str] = None
@@ -149,6 +149,22 @@ class _TopkHW(torch.autograd.Function):
 def topk_hw(probs: torch.Tensor, T: int) -> Tuple[torch.Tensor, torch.Tensor]:
     """[B, HW, P] -> (values [B, P, T] desc-sorted, indices [B, P, T])."""
     return _TopkHW.apply(probs, T)
+
+
+def enqueue_candidates(feat: torch.Tensor, top1_idx: torch.Tensor,
+                       gt: torch.Tensor, C: int, K: int, HW: int):
+    """Per-sample dedup of the GT class's top-1 patches.
+
+    Opt-in HIP path (MGPROTO_HIP_ENQUEUE=1, pending GPU validation);
+    default = the batched torch sort/mask (reference.py), same output."""
+    if (feat.is_cuda and os.environ.get('MGPROTO_HIP_ENQUEUE') == '1'
+            and gt.shape[0] <= 1024 and K <= 32
+            and _load_extension() is not None):
+        ext = _load_extension()
+        rows, lab = ext.enqueue_rows(top1_idx.contiguous(),
+                                     gt.contiguous(), C, K, HW)
+        return feat.index_select(0, rows), lab
+    return reference.enqueue_candidates(feat, top1_idx, gt, C, K, HW)
 
 
 def _em_hip_usable(x: torch.Tensor, K: int) -> bool:

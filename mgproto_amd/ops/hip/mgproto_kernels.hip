@@ -551,6 +551,13 @@ std::vector<torch::Tensor> em_mstep(torch::Tensor x, torch::Tensor logresp,
                                     torch::Tensor means, torch::Tensor covs,
                                     double alpha, double lamda, double eps);
 
+// memory-bank enqueue kernels (enqueue_kernels.hip)
+std::vector<torch::Tensor> enqueue_rows(torch::Tensor top1, torch::Tensor gt,
+                                        int64_t C, int64_t K, int64_t HW);
+void bank_push(torch::Tensor feats, torch::Tensor labels, torch::Tensor mem,
+               torch::Tensor head, torch::Tensor mem_len,
+               int64_t C, int64_t cap);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gmm_fwd", &gmm_fwd, "fused GMM score GEMM forward (MFMA f32)");
     m.def("gmm_bwd", &gmm_bwd, "GMM score GEMM backward wrt features");
@@ -560,4 +567,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bn_bwd", &bn_bwd, "fused BatchNorm(+Add)(+ReLU) backward, NHWC bf16");
     m.def("em_estep", &em_estep, "batched EM e-step (wlp + log-resp)");
     m.def("em_mstep", &em_mstep, "batched EM m-step (closed-form grads + pi)");
+    m.def("enqueue_rows", &enqueue_rows,
+          "per-sample dedup of GT-class top-1 patches -> (rows, labels)");
+    m.def("bank_push", &bank_push,
+          "class-segregated deterministic FIFO ring write");
 }

@@ -21,6 +21,7 @@ cap items (deterministic; unreachable with default shapes since
 B*K <= cap).
 """
 
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -75,6 +76,23 @@ class MemoryBank(nn.Module):
         cap = self.cap_cls
         C = self.num_classes
         dev = feature.device
+
+        if (feature.is_cuda and feature.dtype == torch.float32
+                and os.environ.get('MGPROTO_HIP_ENQUEUE') == '1'):
+            # K5 HIP path (opt-in, pending GPU validation): deterministic
+            # class-segregated ring write, same bank state as the torch
+            # sort/scan/scatter below
+            try:
+                from ..ops import hip_loader
+                ext = hip_loader.load()
+            except Exception:  # noqa: BLE001
+                ext = None
+            if ext is not None:
+                ext.bank_push(feature.contiguous(),
+                              label.to(torch.int64).contiguous(),
+                              self._mem_flat, self.head, self.mem_len,
+                              C, cap)
+                return
 
         order = torch.argsort(label, stable=True)
         lab = label[order]                    # sentinel rows sort last
