@@ -18,7 +18,10 @@ echo "bn-mask tests: $?" >> gpurun_out/r2/summary.txt
 timeout 600 bash -c 'MGPROTO_HIP_EM=1 python -m pytest tests/test_em_hip_gpu.py -x -q' \
     > gpurun_out/r2/em_hip_tests.log 2>&1
 echo "em-hip tests: $?" >> gpurun_out/r2/summary.txt
-timeout 600 bash -c 'MGPROTO_HIP_EM=1 python bench.py --steps 30 --warmup 10' > gpurun_out/r2/bench_emhip.json 2>/dev/null
+timeout 600 bash -c 'MGPROTO_HIP_ENQUEUE=1 python -m pytest tests/test_enqueue_hip_gpu.py -x -q' \
+    > gpurun_out/r2/enqueue_hip_tests.log 2>&1
+echo "enqueue-hip tests: $?" >> gpurun_out/r2/summary.txt
+timeout 600 bash -c 'MGPROTO_HIP_EM=1 MGPROTO_HIP_ENQUEUE=1 python bench.py --steps 30 --warmup 10' > gpurun_out/r2/bench_emhip.json 2>/dev/null
 
 # 3. flagship bench: default, BN-mask on, eager (for the graph delta)
 timeout 600 python bench.py --steps 30 --warmup 10 > gpurun_out/r2/bench_default.json 2>gpurun_out/r2/bench_default.log
