@@ -211,3 +211,36 @@ class TestArgmin:
                 j = int(torch.argmin(dist[b, :, p]))
                 assert idx[b, p] == j
                 assert min_dist[b, p] == dist[b, j, p]
+
+
+class TestDispatchWrappers:
+    def test_cpu_dispatch_equals_reference(self):
+        """ops.em_e_step / em_m_step_grads / enqueue_candidates wrappers
+        route CPU tensors to the oracle regardless of the opt-in envs."""
+        import os
+
+        import mgproto_amd.ops as O
+        os.environ['MGPROTO_HIP_EM'] = '1'
+        os.environ['MGPROTO_HIP_ENQUEUE'] = '1'
+        try:
+            g = torch.Generator().manual_seed(0)
+            x = F.normalize(torch.randn(2, 20, 8, generator=g), dim=2)
+            means = F.normalize(torch.rand(2, 3, 8, generator=g), dim=2)
+            covs = torch.full((2, 3, 8), 0.4)
+            pi = torch.softmax(torch.rand(2, 3, generator=g), dim=1)
+            for a, b in zip(O.em_e_step(x, means, covs, pi),
+                            R.em_e_step(x, means, covs, pi)):
+                assert torch.equal(a, b)
+            wlp, lr = R.em_e_step(x, means, covs, pi)
+            for a, b in zip(O.em_m_step_grads(x, lr, wlp, means, covs),
+                            R.em_m_step_grads(x, lr, wlp, means, covs)):
+                assert torch.equal(a, b)
+            feat = torch.randn(4 * 6, 8, generator=g)
+            top1 = torch.randint(0, 6, (4, 9), generator=g)
+            gt = torch.randint(0, 3, (4,), generator=g)
+            for a, b in zip(O.enqueue_candidates(feat, top1, gt, 3, 3, 6),
+                            R.enqueue_candidates(feat, top1, gt, 3, 3, 6)):
+                assert torch.equal(a, b)
+        finally:
+            os.environ.pop('MGPROTO_HIP_EM', None)
+            os.environ.pop('MGPROTO_HIP_ENQUEUE', None)
