@@ -31,10 +31,13 @@ def _check_contract(rec, n_gpus):
     assert rec['vs_baseline'] is None
     assert rec['data'] == 'synthetic'
     assert rec['value'] > 0 and rec['ms_per_step'] > 0
-    # whole-job value: images/s * ms_per_step ~= global batch images
-    # (both fields are rounded in the output -> 1% relative tolerance)
+    # whole-job value: images/s * ms_per_step ~= global batch images.
+    # Both fields are rounded (2/3 decimals); at very low throughput (e.g.
+    # CPU contention under pytest-xdist) the 0.005 img/s quantum dominates,
+    # so scale the tolerance with it.
     per_step = rec['value'] * rec['ms_per_step'] / 1000.0
-    assert abs(per_step / rec['config']['global_batch'] - 1) < 0.01
+    tol = 0.01 + 0.005 / rec['value'] + 0.0005 / rec['ms_per_step']
+    assert abs(per_step / rec['config']['global_batch'] - 1) < tol
     assert rec['config']['parallelism'] == f'dp{n_gpus}'
 
 
