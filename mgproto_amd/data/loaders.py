@@ -9,6 +9,7 @@ from typing import Tuple
 
 import torch
 from torch.utils.data import DataLoader, Subset
+from torch.utils.data.distributed import DistributedSampler
 
 from . import transforms as T
 from .folder import ImageFolder
@@ -27,13 +28,32 @@ def _shard(ds, world, rank):
     return ds
 
 
+def make_train_sampler(ds, world: int, rank: int, seed: int = 0):
+    """Globally-reshuffled, padded train sharding.
+
+    Every epoch all ranks draw from ONE global permutation (call
+    ``sampler.set_epoch(epoch)`` — the trainer does), and shards are padded
+    to equal length so every rank issues the same number of gradient
+    all-reduces per epoch. A static ``range(rank, N, world)`` shard would
+    (a) freeze which samples a rank ever sees — the reference's
+    DataParallel saw globally shuffled batches every epoch
+    (reference main.py:96-104) — and (b) let per-rank batch counts differ
+    by one, desynchronizing the reducer's RCCL collectives."""
+    if world <= 1:
+        return None
+    return DistributedSampler(ds, num_replicas=world, rank=rank,
+                              shuffle=True, seed=seed, drop_last=False)
+
+
 def build_image_loaders(cfg, world: int = 1, rank: int = 0,
-                        fast_augment: bool = True,
+                        fast_augment: bool = None,
                         scaled_decode: bool = None) -> Tuple:
     img_size = cfg.img_size
+    import os
     if scaled_decode is None:
-        import os
         scaled_decode = os.environ.get('MGPROTO_SCALED_DECODE', '0') == '1'
+    if fast_augment is None:
+        fast_augment = getattr(cfg, 'fast_augment', True)
     normalize = T.Normalize(mean=mean, std=std)
 
     if fast_augment:
@@ -75,17 +95,22 @@ def build_image_loaders(cfg, world: int = 1, rank: int = 0,
     push_ds = ImageFolder(cfg.train_push_dir, push_tf)
     test_ds = ImageFolder(cfg.test_dir, test_tf)
 
-    def mk(ds, bs, shuffle=False):
-        return DataLoader(ds, batch_size=bs, shuffle=shuffle,
+    def mk(ds, bs, shuffle=False, sampler=None):
+        return DataLoader(ds, batch_size=bs,
+                          shuffle=shuffle if sampler is None else False,
+                          sampler=sampler,
                           num_workers=cfg.num_workers, pin_memory=True,
                           collate_fn=_collate, persistent_workers=cfg.num_workers > 0)
 
-    train_loader = mk(_shard(train_ds, world, rank), cfg.train_batch_size,
-                      shuffle=True)
+    train_sampler = make_train_sampler(train_ds, world, rank)
+    train_loader = mk(train_ds, cfg.train_batch_size,
+                      shuffle=True, sampler=train_sampler)
+    # push/test shards stay static stride shards: they run under no_grad
+    # with a single post-loop collective, so uneven batch counts are safe,
+    # and padding would double-count images in the metrics
     push_loader = mk(_shard(push_ds, world, rank), cfg.train_push_batch_size)
     test_loader = mk(_shard(test_ds, world, rank), cfg.test_batch_size)
 
-    import os
     ood_loaders = []
     for d in (cfg.test_dir_ood1, cfg.test_dir_ood2):
         if os.path.isdir(d):

@@ -229,6 +229,10 @@ class MGProto(nn.Module):
                                                      dtype=torch.int64),
                              persistent=False)
         self.prototype_optimizer = None  # API-compat slot; unused internally
+        # NOTE: the _em_* Adam buffers are non-persistent so the exported
+        # state_dict stays exactly the reference checkpoint layout
+        # (tests/test_checkpoint_interop.py); the training resume path
+        # carries them separately via em_state_dict()/load_em_state().
 
         # pluggable enqueue: the distributed wrapper replaces this with an
         # all-gather + replicated push (parallel/state_sync.py)
@@ -324,6 +328,26 @@ class MGProto(nn.Module):
         distances = -probs.view(B, H * W, self.num_prototypes) \
                           .permute(0, 2, 1).reshape(B, self.num_prototypes, H, W)
         return base_feature, distances
+
+    def em_state_dict(self) -> dict:
+        """Per-class EM Adam state for full resume. Kept OUT of the module
+        state_dict so checkpoints stay byte-compatible with the reference
+        layout; a resumed run restores it via load_em_state so the
+        prototype trajectory continues instead of restarting from zeroed
+        moments (the reference's external prototype optimizer state was
+        never saved at all, reference utils/save.py:5-12)."""
+        return {'exp_avg': self._em_exp_avg.detach().cpu().clone(),
+                'exp_avg_sq': self._em_exp_avg_sq.detach().cpu().clone(),
+                'step': self._em_step.detach().cpu().clone()}
+
+    def load_em_state(self, state: dict):
+        if not state:
+            return
+        with torch.no_grad():
+            self._em_exp_avg.copy_(state['exp_avg'].to(self._em_exp_avg.device))
+            self._em_exp_avg_sq.copy_(
+                state['exp_avg_sq'].to(self._em_exp_avg_sq.device))
+            self._em_step.copy_(state['step'].to(self._em_step.device))
 
     # -------------------------------------------------------------------- EM
     @torch.no_grad()

@@ -45,7 +45,11 @@ class Comm:
         self.backend = backend
         if self.world_size > 1 and not dist.is_initialized():
             if backend is None:
-                backend = 'nccl' if self.device.type == 'cuda' else 'gloo'
+                # register gloo alongside RCCL so the occasional CPU-tensor
+                # collective (checkpoint metadata, host-side merges) is
+                # routed to gloo instead of crashing the nccl group
+                backend = ('cpu:gloo,cuda:nccl' if self.device.type == 'cuda'
+                           else 'gloo')
             self.backend = backend
             if self.device.type == 'cuda':
                 torch.cuda.set_device(self.device)

@@ -29,9 +29,25 @@ class BucketedGradReducer:
         self.bucket_bytes = bucket_mb * 1024 * 1024
         self._hooks = []
         self._works: List = []
+        self._modules = module if isinstance(module, (list, tuple)) else [module]
+        self._skip = set(id(p) for p in skip_params)
+        self._build()
 
-        skip = set(id(p) for p in skip_params)
-        modules = module if isinstance(module, (list, tuple)) else [module]
+    def rebuild(self):
+        """Re-derive bucket membership from the CURRENT requires_grad flags.
+
+        warm_only()/joint() freeze and unfreeze the backbone mid-run
+        (engine/trainer.py); without a rebuild, buckets mixing frozen and
+        live params never fill during backward and fall back to
+        synchronous reduces in finalize() — correct but unoverlapped.
+        Call after every phase flip (train.py does)."""
+        self.remove()
+        self._works.clear()
+        self._build()
+
+    def _build(self):
+        skip = self._skip
+        modules = self._modules
         params = [p for m in modules for p in m.parameters()
                   if p.requires_grad and id(p) not in skip]
         # reverse order ~ backward completion order for sequential nets

@@ -75,7 +75,12 @@ def gather_push_candidates(comm: Comm, dists: torch.Tensor,
     """
     if not comm.is_distributed:
         return dists, meta
-    flat_meta = meta.reshape(-1).to(torch.float64)
-    d_all = comm.all_gather_varlen(dists.double())
+    # collectives must run on the backend's device: push.py collects the
+    # candidates on the host, but an RCCL process group rejects CPU
+    # tensors — stage through comm.device and hand CPU results back to
+    # the (host-side) merge
+    dev = comm.device
+    flat_meta = meta.reshape(-1).to(device=dev, dtype=torch.float64)
+    d_all = comm.all_gather_varlen(dists.to(dev).double())
     m_all = comm.all_gather_varlen(flat_meta)
-    return d_all.float(), m_all.view(-1, 4).to(torch.int64)
+    return d_all.float().cpu(), m_all.view(-1, 4).to(torch.int64).cpu()

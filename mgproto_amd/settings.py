@@ -98,6 +98,10 @@ class Settings:
     test_batch_size: int = test_batch_size
     train_push_batch_size: int = train_push_batch_size
     num_workers: int = 8  # the reference ran num_workers=0 (main.py:94); we don't
+    # True = one-homography fused augmentation (fast, ~4x loader throughput,
+    # approximates the reference's 4-pass chain); False = the faithful
+    # reference chain (main.py:98-104) for training-parity comparisons
+    fast_augment: bool = True
 
     # optimization
     joint_optimizer_lrs: Dict[str, float] = field(default_factory=lambda: dict(joint_optimizer_lrs))

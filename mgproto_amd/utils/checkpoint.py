@@ -40,6 +40,9 @@ def save_train_state(path: str, model, optimizers: Dict[str, object],
     import numpy as np
     state = {
         'model': model.state_dict(),
+        # EM Adam moments live outside the (reference-layout) state_dict
+        'em_state': (model.em_state_dict()
+                     if hasattr(model, 'em_state_dict') else None),
         'optimizers': {k: v.state_dict() for k, v in optimizers.items() if v is not None},
         'schedulers': {k: v.state_dict() for k, v in schedulers.items() if v is not None},
         'epoch': epoch,
@@ -60,6 +63,8 @@ def load_train_state(path: str, model, optimizers: Dict[str, object] = None,
                      map_location='cpu') -> dict:
     state = torch.load(path, map_location=map_location, weights_only=False)
     model.load_state_dict(state['model'])
+    if state.get('em_state') is not None and hasattr(model, 'load_em_state'):
+        model.load_em_state(state['em_state'])
     for k, v in (optimizers or {}).items():
         if v is not None and k in state.get('optimizers', {}):
             v.load_state_dict(state['optimizers'][k])

@@ -64,9 +64,12 @@ def build_loaders(cfg: Settings, comm):
                 torch.tensor([b[1] for b in batch]),
                 torch.tensor([b[2] for b in batch]))
 
-    mk = lambda ds, bs: DataLoader(ds, batch_size=bs, num_workers=cfg.num_workers,  # noqa: E731
-                                   collate_fn=collate, drop_last=False)
-    return (mk(shard(train_ds), cfg.train_batch_size),
+    mk = lambda ds, bs, sampler=None: DataLoader(  # noqa: E731
+        ds, batch_size=bs, num_workers=cfg.num_workers,
+        collate_fn=collate, drop_last=False, sampler=sampler)
+    from mgproto_amd.data.loaders import make_train_sampler
+    train_sampler = make_train_sampler(train_ds, world, rank)
+    return (mk(train_ds, cfg.train_batch_size, sampler=train_sampler),
             mk(shard(push_ds), cfg.train_push_batch_size),
             mk(shard(test_ds), cfg.test_batch_size),
             mk(shard(ood1), cfg.test_batch_size),
@@ -99,11 +102,17 @@ def main():
     parser.add_argument('--seed', type=int, default=None,
                         help='deterministic init/data seeding (the reference '
                              'ships seeding commented out, main.py:45-49)')
+    parser.add_argument('--faithful-aug', action='store_true',
+                        help='use the reference\'s exact 4-pass augmentation '
+                             'chain (main.py:98-104) instead of the fused '
+                             'one-homography approximation — pick this for '
+                             'training-parity comparisons')
     args = parser.parse_args()
 
     cfg = Settings(base_architecture=args.arch, aux_loss=args.aux_loss,
                    sz_embedding=args.aux_emb_sz, mem_capacity=args.mem_sz,
-                   mine_K=args.mine_level)
+                   mine_K=args.mine_level,
+                   fast_augment=not args.faithful_aug)
     if args.addon:
         cfg.add_on_layers_type = args.addon
     if os.environ.get('MGPROTO_TINY_TEST') == '1':
@@ -248,8 +257,15 @@ def main():
     log('start training')
     decay_epochs = cfg.lr_decay_epochs()
     epoch = start_epoch
+    last_phase = None
     for epoch in range(start_epoch, cfg.num_train_epochs):
         _save_now.epoch = epoch
+        # global per-epoch reshuffle (DistributedSampler contract); the
+        # prefetcher wrapper exposes the underlying loader's sampler
+        sampler = getattr(getattr(train_loader, 'loader', train_loader),
+                          'sampler', None)
+        if hasattr(sampler, 'set_epoch'):
+            sampler.set_epoch(epoch)
         log('epoch: \t{0}'.format(epoch))
         use_mining = epoch >= cfg.mine_start
         update_GMM = (epoch >= cfg.updateGMM_start
@@ -259,20 +275,24 @@ def main():
 
         kw = dict(device=device, amp_dtype=cfg.amp_dtype, comm=comm,
                   metrics=metrics)
-        if epoch < cfg.num_warm_epochs:
+        phase = 'warm' if epoch < cfg.num_warm_epochs else 'joint'
+        if phase == 'warm':
             tnt.warm_only(ppnet, log=log)
-            _, train_results = tnt.train(
-                ppnet, train_loader, warm_optimizer, aux_criterion=aux_criterion,
-                use_mine=use_mining, update_GMM=update_GMM, coefs=cfg.coefs,
-                log=log, em_runner=em_runner, reducer=reducer, **kw)
+            optimizer = warm_optimizer
         else:
             tnt.joint(ppnet, log=log)
             if epoch in decay_epochs:
                 joint_lr_scheduler.step()
-            _, train_results = tnt.train(
-                ppnet, train_loader, joint_optimizer, aux_criterion=aux_criterion,
-                use_mine=use_mining, update_GMM=update_GMM, coefs=cfg.coefs,
-                log=log, em_runner=em_runner, reducer=reducer, **kw)
+            optimizer = joint_optimizer
+        if reducer is not None and phase != last_phase:
+            # bucket membership follows requires_grad: rebuild on phase
+            # flips so warm-phase reduces overlap with backward too
+            reducer.rebuild()
+        last_phase = phase
+        _, train_results = tnt.train(
+            ppnet, train_loader, optimizer, aux_criterion=aux_criterion,
+            use_mine=use_mining, update_GMM=update_GMM, coefs=cfg.coefs,
+            log=log, em_runner=em_runner, reducer=reducer, **kw)
 
         if args.ood_eval:
             accu, _ = tnt.test(ppnet, (test_loader, ood1_loader, ood2_loader),
