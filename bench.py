@@ -175,30 +175,26 @@ def main():
     model.train()
     graph = None
     if use_graph:
-        static_img = pool.images[0].clone()
-        static_tgt = pool.labels[0].clone()
-        try:
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                for _ in range(3):
-                    compute(static_img, static_tgt)
-            torch.cuda.current_stream().wait_stream(side)
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                compute(static_img, static_tgt)
-        except Exception as e:  # noqa: BLE001
-            print(f'# hipGraph capture failed ({type(e).__name__}: {e}); '
-                  'falling back to eager', flush=True)
-            graph = None
-            use_graph = False
+        # the SAME whole-step capture train.py runs (engine/graphstep.py) —
+        # the bench measures the production path, not a bench-only one
+        from mgproto_amd.engine.graphstep import GraphedStep
+        gs = GraphedStep(model, aux, coefs, device, args.batch, args.img,
+                         amp_dtype='bf16', channels_last=True)
 
-    if graph is not None:
         def step():  # noqa: F811
             image, target = pool.next()
-            static_img.copy_(image)
-            static_tgt.copy_(target)
-            graph.replay()
+            if gs.matches(image, opt, True, use_em):
+                gs.step(image, target, opt, reducer=reducer,
+                        em_active=use_em, use_mine=True)
+            else:
+                compute(image, target)
+        # prime: eager warmups + capture happen inside gs.step; make sure
+        # they are done before the timed region regardless of --warmup
+        for _ in range(gs.warmup_steps + 1):
+            step()
+        graph = gs.graph
+        if graph is None:
+            use_graph = False
 
     for _ in range(args.warmup):
         step()

@@ -69,7 +69,7 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
               use_mine=False, update_GMM=False, class_specific=True,
               coefs=None, log=print, device=None, amp_dtype='bf16',
               em_runner: Optional[EMRunner] = None, metrics=None,
-              comm=None, print_every=20, reducer=None):
+              comm=None, print_every=20, reducer=None, graph_step=None):
     device = device or next(_unwrap(model).parameters()).device
     m = _unwrap(model)
     start = time.time()
@@ -98,49 +98,73 @@ def _training(model, dataloader, optimizer=None, aux_criterion=None,
         if em_runner is not None:
             em_runner.sync()  # next forward reads EM-updated means/priors
 
-        with timer.phase('forward'), _amp_ctx(device, amp_dtype):
-            output, x_auxiliary = model(image, target)
+        # hipGraph fast path: full-size batch under a stable step program
+        # replays the captured whole-step graph (engine/graphstep.py);
+        # remainder batches and signature flips fall through to eager
+        em_now = (update_GMM
+                  and (mem_nonempty or (i == 0 and target.numel() > 0))
+                  and (iter_base + i + 1) % m.update_interval == 0)
+        if (graph_step is not None
+                and (not update_GMM or m.update_interval == 1)
+                and graph_step.matches(image, optimizer, use_mine, em_now)):
+            with timer.phase('forward'):
+                st = graph_step.step(image, target, optimizer,
+                                     reducer=reducer, em_active=em_now,
+                                     use_mine=use_mine)
+            loss, cross_entropy = st['loss'], st['ce']
+            mine_loss, aux_loss = st['mine'], st['aux']
+            n_examples += target.numel()
+            n_correct += st['n_correct']
+            n_batches += 1
+            total_ce += st['ce']
+            total_mine += st['mine']
+            total_aux += st['aux']
+            if update_GMM:
+                mem_nonempty = mem_nonempty or (i == 0 and target.numel() > 0)
+        else:
+            with timer.phase('forward'), _amp_ctx(device, amp_dtype):
+                output, x_auxiliary = model(image, target)
 
-        with timer.phase('loss'):
-            output = output.float()
-            if use_mine and output.shape[2] > 1:
-                mine_loss = sum(F.cross_entropy(output[:, :, k], target)
-                                for k in range(1, output.shape[2])) \
-                    / (output.shape[2] - 1)
-            else:
-                mine_loss = torch.zeros((), device=device)
-            cross_entropy = F.cross_entropy(output[:, :, 0], target)
-            aux_loss = (aux_criterion(x_auxiliary.float(), target)
-                        if aux_criterion is not None
-                        else torch.zeros((), device=device))
+            with timer.phase('loss'):
+                output = output.float()
+                if use_mine and output.shape[2] > 1:
+                    mine_loss = sum(F.cross_entropy(output[:, :, k], target)
+                                    for k in range(1, output.shape[2])) \
+                        / (output.shape[2] - 1)
+                else:
+                    mine_loss = torch.zeros((), device=device)
+                cross_entropy = F.cross_entropy(output[:, :, 0], target)
+                aux_loss = (aux_criterion(x_auxiliary.float(), target)
+                            if aux_criterion is not None
+                            else torch.zeros((), device=device))
 
-        predicted = torch.argmax(output[:, :, 0].detach(), dim=1)
-        n_examples += target.numel()
-        n_correct += (predicted == target).sum()
-        n_batches += 1
-        total_ce += cross_entropy.detach()
-        total_mine += mine_loss.detach()
-        total_aux += aux_loss.detach()
+            predicted = torch.argmax(output[:, :, 0].detach(), dim=1)
+            n_examples += target.numel()
+            n_correct += (predicted == target).sum()
+            n_batches += 1
+            total_ce += cross_entropy.detach()
+            total_mine += mine_loss.detach()
+            total_aux += aux_loss.detach()
 
-        loss = (coefs['crs_ent'] * cross_entropy + coefs['mine'] * mine_loss
-                + coefs['aux'] * aux_loss)
-        with timer.phase('backward'):
-            if reducer is not None:
-                reducer.prepare()   # arm per-step bucket state (C1)
-            optimizer.zero_grad(set_to_none=True)
-            loss.backward()
-        with timer.phase('optimizer'):
-            if reducer is not None:
-                reducer.finalize()  # drain async all-reduces before step
-            optimizer.step()
+            loss = (coefs['crs_ent'] * cross_entropy + coefs['mine'] * mine_loss
+                    + coefs['aux'] * aux_loss)
+            with timer.phase('backward'):
+                if reducer is not None:
+                    reducer.prepare()   # arm per-step bucket state (C1)
+                optimizer.zero_grad(set_to_none=True)
+                loss.backward()
+            with timer.phase('optimizer'):
+                if reducer is not None:
+                    reducer.finalize()  # drain async all-reduces before step
+                optimizer.step()
 
-        # EM update (reference train_and_test.py:61-63; update_interval=1)
-        if update_GMM:
-            mem_nonempty = mem_nonempty or (i == 0 and target.numel() > 0)
-            if mem_nonempty and (iter_base + i + 1) % m.update_interval == 0:
-                with timer.phase('em'):
-                    (em_runner.run() if em_runner is not None
-                     else m.update_GMM())
+            # EM update (reference train_and_test.py:61-63; update_interval=1)
+            if update_GMM:
+                mem_nonempty = mem_nonempty or (i == 0 and target.numel() > 0)
+                if mem_nonempty and (iter_base + i + 1) % m.update_interval == 0:
+                    with timer.phase('em'):
+                        (em_runner.run() if em_runner is not None
+                         else m.update_GMM())
 
         if print_every and i % print_every == 0 \
                 and (comm is None or comm.rank == 0):

@@ -123,6 +123,7 @@ class Settings:
     amp_dtype: str = 'bf16'          # backbone autocast dtype; prototype math stays fp32
     channels_last: bool = True       # NHWC layout for MIOpen convs
     em_stream: bool = True           # run the EM update on a side HIP stream
+    hip_graph: bool = True           # capture the whole training step in a hipGraph
     grad_bucket_mb: int = 50         # RCCL all-reduce bucket size (xGMI-tuned)
     native_ops: str = 'auto'         # 'auto' | 'force' | 'off' — HIP kernel dispatch
 

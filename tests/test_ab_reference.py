@@ -1,0 +1,68 @@
+"""End-to-end learning equivalence vs the ORIGINAL reference code.
+
+tools/ab_reference.py imports the reference implementation from
+/root/reference and trains it next to mgproto_amd on identical weights,
+batches and schedule (joint phase, mining + enqueue + EM active). This
+asserts the trajectories match — op-level parity tests can all pass while
+the composed dynamics diverge (wrong EM trigger order, enqueue dedup
+semantics, mask application); this test catches that class of bug.
+"""
+
+import os
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), '..'))
+
+if not os.path.isdir('/root/reference'):
+    pytest.skip('reference checkout not present', allow_module_level=True)
+
+
+@pytest.fixture(scope='module')
+def ab_trajs():
+    from tools.ab_reference import run_ab
+    torch.manual_seed(0)
+    return run_ab(n_steps=12, batch=16)
+
+
+def test_first_step_losses_identical(ab_trajs):
+    """Identical weights + inputs -> step-0 forward/backward must agree to
+    float-accumulation noise (this is composed-forward parity)."""
+    ref, ours = ab_trajs
+    assert abs(ref[0]['loss'] - ours[0]['loss']) < 1e-3, \
+        (ref[0]['loss'], ours[0]['loss'])
+    assert abs(ref[0]['ce'] - ours[0]['ce']) < 1e-4
+    assert abs(ref[0]['mine'] - ours[0]['mine']) < 1e-4
+    assert abs(ref[0]['aux'] - ours[0]['aux']) < 1e-3
+    assert ref[0]['acc'] == ours[0]['acc']
+
+
+def test_early_trajectory_tracks(ab_trajs):
+    """First few optimizer+EM steps stay within tight drift bounds."""
+    ref, ours = ab_trajs
+    for i in range(4):
+        assert abs(ref[i]['loss'] - ours[i]['loss']) < 0.08, \
+            (i, ref[i]['loss'], ours[i]['loss'])
+        assert abs(ref[i]['ce'] - ours[i]['ce']) < 0.05, (i,)
+
+
+def test_full_trajectory_within_tolerance(ab_trajs):
+    """Over 12 steps (EM every step) the trajectories may drift by float
+    chaos but must stay on the same curve and both must be learning."""
+    ref, ours = ab_trajs
+    for i, (r, o) in enumerate(zip(ref, ours)):
+        assert abs(r['ce'] - o['ce']) < 0.3, (i, r['ce'], o['ce'])
+    # both converging on the separable task
+    assert ref[-1]['ce'] < ref[0]['ce']
+    assert ours[-1]['ce'] < ours[0]['ce']
+    assert abs(ref[-1]['acc'] - ours[-1]['acc']) <= 0.25
+
+
+def test_reference_state_dict_loads_strict(ab_trajs):
+    """run_ab already loaded a REAL reference-produced state_dict into our
+    model with strict=True (tools/ab_reference.py run_ab) — reaching here
+    means the interop contract held against actual reference tensors, not
+    just our own export."""
+    assert ab_trajs is not None

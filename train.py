@@ -190,23 +190,43 @@ def main():
                                    sz_embed=args.aux_emb_sz, mrg=0.1,
                                    beta=32).to(device)
 
-    joint_optimizer = torch.optim.Adam([
+    from mgproto_amd.engine.graphstep import GraphedStep, graphs_enabled
+    world = comm.world_size if comm else 1
+    use_graph = cfg.hip_graph and graphs_enabled(device, world)
+
+    def _adam(groups, capturable):
+        # fused+capturable Adam keeps the whole optimizer step on device
+        # (required inside a hipGraph; harmless outside one)
+        try:
+            return torch.optim.Adam(groups, fused=True, capturable=capturable)
+        except (RuntimeError, TypeError, ValueError):
+            return torch.optim.Adam(groups, capturable=capturable)
+
+    joint_optimizer = _adam([
         {'params': ppnet.features.parameters(),
          'lr': cfg.joint_optimizer_lrs['features'], 'weight_decay': 1e-4},
         {'params': ppnet.add_on_layers.parameters(),
          'lr': cfg.joint_optimizer_lrs['add_on_layers'], 'weight_decay': 1e-4},
         {'params': aux_criterion.parameters(),
          'lr': cfg.joint_optimizer_lrs['features'] * 100, 'weight_decay': 1e-4},
-    ])
+    ], use_graph)
     joint_lr_scheduler = torch.optim.lr_scheduler.StepLR(
         joint_optimizer, step_size=1, gamma=cfg.joint_lr_gamma)
-    warm_optimizer = torch.optim.Adam([
+    warm_optimizer = _adam([
         {'params': ppnet.add_on_layers.parameters(),
          'lr': cfg.warm_optimizer_lrs['add_on_layers'], 'weight_decay': 1e-4},
         {'params': aux_criterion.parameters(),
          'lr': cfg.joint_optimizer_lrs['features'] * 100, 'weight_decay': 1e-4},
-    ])
+    ], use_graph)
     ppnet.prototype_lr = cfg.joint_optimizer_lrs['prototype_vectors']
+
+    graph_step = None
+    if use_graph:
+        graph_step = GraphedStep(ppnet, aux_criterion, cfg.coefs, device,
+                                 cfg.train_batch_size, cfg.img_size,
+                                 amp_dtype=cfg.amp_dtype,
+                                 channels_last=cfg.channels_last)
+        log('hipGraph training step: enabled')
 
     reducer = None
     if comm is not None and comm.is_distributed:
@@ -292,7 +312,8 @@ def main():
         _, train_results = tnt.train(
             ppnet, train_loader, optimizer, aux_criterion=aux_criterion,
             use_mine=use_mining, update_GMM=update_GMM, coefs=cfg.coefs,
-            log=log, em_runner=em_runner, reducer=reducer, **kw)
+            log=log, em_runner=em_runner, reducer=reducer,
+            graph_step=graph_step, **kw)
 
         if args.ood_eval:
             accu, _ = tnt.test(ppnet, (test_loader, ood1_loader, ood2_loader),
