@@ -25,8 +25,8 @@ import torch.nn.functional as F
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument('--gpus', type=int, default=1)
-    ap.add_argument('--steps', type=int, default=20)
-    ap.add_argument('--warmup', type=int, default=5)
+    ap.add_argument('--steps', type=int, default=100)
+    ap.add_argument('--warmup', type=int, default=10)
     ap.add_argument('--batch', type=int, default=80)
     ap.add_argument('--arch', type=str, default='resnet50')
     ap.add_argument('--addon', type=str, default='regular_upsample',

@@ -43,11 +43,14 @@ def _usable(x: torch.Tensor, residual) -> bool:
 
 
 def _use_mask() -> bool:
-    """EXPERIMENTAL (round 2, MGPROTO_BN_MASK=1): the forward emits a 1-bit
-    relu mask per element so the backward never re-reads y (saves ~2 bf16
-    activation passes of backward traffic). Bit-identical semantics to the
-    y>0 test; pending GPU validation."""
-    return os.environ.get('MGPROTO_BN_MASK') == '1'
+    """Forward emits a 1-bit relu mask per element so the backward never
+    re-reads y (saves ~2 bf16 activation passes of backward traffic).
+
+    DEFAULT ON since round 2: parity-green on MI355X (20/20
+    tests/test_fused_bn_gpu.py with MGPROTO_BN_MASK=1) and faster
+    end-to-end (1737.9 vs 1690.6 img/s A/B, gpurun_out/r2). Set
+    MGPROTO_BN_MASK=0 to fall back to the y>0 re-read backward."""
+    return os.environ.get('MGPROTO_BN_MASK', '1') == '1'
 
 
 class _FusedBN(torch.autograd.Function):

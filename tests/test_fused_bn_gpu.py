@@ -179,13 +179,11 @@ def test_gemm_conv1x1_parity_optin():
         os.environ.pop('MGPROTO_GEMM_CONV1X1')
 
 
-@pytest.mark.skipif(os.environ.get('MGPROTO_BN_MASK') != '1',
-                    reason='experimental relu-mask BN path (round 2): '
-                           'set MGPROTO_BN_MASK=1 to validate')
 @pytest.mark.parametrize('res', [False, True])
 def test_bn_mask_backward_matches_y_path(res):
-    """EXPERIMENTAL: mask-based backward must be bitwise-identical to the
-    y-read backward (the mask is derived from the same rounded y)."""
+    """Mask-based backward (default since round 2) must be bitwise-identical
+    to the y-read backward (the mask is derived from the same rounded y).
+    The test drives both paths itself via the env var."""
     from mgproto_amd.models.fused_bn import bn_act
     torch.manual_seed(3)
     C = 64
