@@ -58,31 +58,58 @@ def test_bank_push_matches_torch_path(seed):
 
 
 def test_forward_enqueue_matches_torch_path():
-    """Full forward with HIP enqueue+push == torch path (same bank)."""
+    """In-situ parity: drive BOTH enqueue paths from the SAME forward's
+    tensors (the real strided top-1 slice, real features) and compare the
+    resulting bank states bit-for-bit.
+
+    Round-2 note: the original form of this test built two models and
+    compared their banks after separate forwards; the round-2 debug run
+    (gpurun_out/r2b/enqueue_debug.log) showed the FEATURES already differ
+    between two same-seed model instances in one process (MIOpen algo
+    timing jitter upstream), so cross-run bitwise bank equality was an
+    invalid oracle — the kernels themselves are bit-identical on the same
+    inputs, which is what this asserts."""
+    import mgproto_amd.ops as O
     from mgproto_amd.model import construct_MGProto
 
-    def run(enable):
-        os.environ['MGPROTO_HIP_ENQUEUE'] = '1' if enable else '0'
-        torch.manual_seed(0)
-        C, K, d = 5, 3, 16
-        m = construct_MGProto('resnet18', pretrained=False, img_size=64,
-                              prototype_shape=(C * K, d, 1, 1),
-                              num_classes=C, add_on_layers_type='regular',
-                              sz_embedding=8, mem_capacity=8,
-                              mine_K=2).cuda()
+    os.environ['MGPROTO_HIP_ENQUEUE'] = '1'
+    torch.manual_seed(0)
+    C, K, d = 5, 3, 16
+    m = construct_MGProto('resnet18', pretrained=False, img_size=64,
+                          prototype_shape=(C * K, d, 1, 1),
+                          num_classes=C, add_on_layers_type='regular',
+                          sz_embedding=8, mem_capacity=8,
+                          mine_K=2).cuda()
+    captured = []
+    orig = O.enqueue_candidates
+
+    def spy(feat, top1, gt, C_, K_, HW_):
+        captured.append((feat, top1, gt, C_, K_, HW_))
+        return orig(feat, top1, gt, C_, K_, HW_)
+
+    import mgproto_amd.model as MM
+    MM.ops.enqueue_candidates = spy
+    try:
         g = torch.Generator().manual_seed(3)
         x = torch.randn(6, 3, 64, 64, generator=g).cuda()
         y = (torch.arange(6) % C).cuda()
         with torch.no_grad():
             m(x, y)
             m(x.flip(0), y.flip(0))
-        return (m.queue.mem.cpu(), m.queue.mem_len.cpu(),
-                m.memory_updated_cls.cpu())
-    try:
-        mem_h, len_h, dirty_h = run(True)
-        mem_r, len_r, dirty_r = run(False)
     finally:
-        os.environ['MGPROTO_HIP_ENQUEUE'] = '1'
-    assert torch.equal(mem_h, mem_r)
-    assert torch.equal(len_h, len_r)
-    assert torch.equal(dirty_h, dirty_r)
+        MM.ops.enqueue_candidates = orig
+    assert len(captured) == 2
+
+    # replay both paths over the captured in-situ tensors
+    banks = {}
+    for env in ('1', '0'):
+        os.environ['MGPROTO_HIP_ENQUEUE'] = env
+        bank = MemoryBank(C, d, capacity=C * 8).cuda()
+        for (feat, top1, gt, C_, K_, HW_) in captured:
+            f, lab = O.enqueue_candidates(feat, top1, gt, C_, K_, HW_)
+            bank.push(f, lab)
+        banks[env] = (bank.mem.cpu(), bank.mem_len.cpu(), bank.head.cpu())
+    os.environ['MGPROTO_HIP_ENQUEUE'] = '1'
+    assert torch.equal(banks['1'][0], banks['0'][0])
+    assert torch.equal(banks['1'][1], banks['0'][1])
+    assert torch.equal(banks['1'][2], banks['0'][2])
