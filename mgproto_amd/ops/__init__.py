@@ -104,16 +104,103 @@ class _GMMScore(torch.autograd.Function):
         return grad_feat, None, None, None
 
 
+class _GMMScoreUniform(torch.autograd.Function):
+    """Reduced GMM score for FROZEN isotropic sigma (the default model:
+    prototype_covs is initialized to 1/sqrt(2 pi) and never updated,
+    reference model.py:151-152). Every x^2-column of W is then the constant
+    cuni = -1/(2 sigma^2), so
+
+        out[n,p] = f(bias[p] + x[n] . A[p] + cuni * ||x[n]||^2)
+
+    — HALF the GEMM of the general form. Exact fp32; falls back to torch
+    off-GPU."""
+
+    @staticmethod
+    def forward(ctx, feat, A, bias, cuni, apply_exp):
+        d = feat.shape[1]
+        rn2 = (feat * feat).sum(dim=1).contiguous()
+        ext = _native_or_die() if feat.is_cuda else None
+        if ext is not None and (d % 8 != 0 or d > 128):
+            ext = None
+        ctx.used_ext = ext is not None
+        if ext is not None:
+            out = ext.gmm_fwd_uni(feat, A, bias, rn2, cuni, apply_exp)
+        else:
+            out = torch.addmm(bias.unsqueeze(0), feat, A.t()) \
+                + cuni * rn2.unsqueeze(1)
+            if apply_exp:
+                out = torch.exp(out)
+        ctx.save_for_backward(feat, A, out if apply_exp else torch.empty(0))
+        ctx.apply_exp = apply_exp
+        ctx.cuni = cuni
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        feat, A, probs = ctx.saved_tensors
+        g = grad_out * probs if ctx.apply_exp else grad_out
+        rs = g.sum(dim=1).contiguous()
+        ext = _native_or_die() if (feat.is_cuda and ctx.used_ext) else None
+        if ext is not None:
+            P = A.shape[0]
+            if P % 4:
+                pad = 4 - P % 4
+                g = torch.nn.functional.pad(g, (0, pad))
+                A = torch.nn.functional.pad(A, (0, 0, 0, pad))
+            grad_feat = ext.gmm_bwd_uni(g.contiguous(), feat,
+                                        A.t().contiguous(), rs, ctx.cuni)
+        else:
+            grad_feat = g @ A + (2.0 * ctx.cuni) * feat * rs.unsqueeze(1)
+        return grad_feat, None, None, None, None
+
+
+# uniform-sigma detection cache, keyed on the MODEL's covs tensor
+# (stable identity across calls, unlike the per-call reshape view): the
+# one host sync happens on first sight of a covs tensor — during an eager
+# warmup step — never inside a hipGraph capture/replay
+_uniform_cache = {}
+
+
+def _uniform_inv_var(covs: torch.Tensor):
+    """inv_var scalar if ALL sigma entries are equal, else None."""
+    key = (id(covs), covs._version, covs.data_ptr(), covs.shape)
+    hit = _uniform_cache.get(key)
+    if hit is not None:
+        return hit[0]
+    flat = covs.detach().reshape(-1).float()
+    first = flat[0]
+    uniform = bool((flat == first).all())
+    val = float(1.0 / (first * first)) if uniform else None
+    if len(_uniform_cache) > 64:     # bound: models come and go in tests
+        _uniform_cache.clear()
+    _uniform_cache[key] = (val,)
+    return val
+
+
 def gmm_scores(feat: torch.Tensor, means: torch.Tensor, covs: torch.Tensor,
                apply_exp: bool = True, eps: float = 0.0) -> torch.Tensor:
     """[N, d] patch features -> [N, P] Gaussian (log-)likelihoods.
 
     The hot K1 op (reference model.py:213-215). ``apply_exp=True`` returns
     probabilities (forward path); ``False`` returns log-probabilities
-    (`compute_log_prob` API parity).
-    """
+    (`compute_log_prob` API parity). Frozen-isotropic covariances (the
+    default model) dispatch to the half-K reduced kernel; a general diag
+    sigma keeps the full [x, x^2] GEMM (checkpoint parity)."""
     means2 = means.reshape(-1, means.shape[-1]).detach()
     covs2 = covs.reshape(-1, covs.shape[-1]).detach()
+    if eps == 0.0 and os.environ.get('MGPROTO_NO_GMM_UNI') != '1':
+        inv_var = _uniform_inv_var(covs)
+        if inv_var is not None:
+            import math
+            m = means2.float()
+            A = (m * inv_var).contiguous()                       # [P, d]
+            d = m.shape[1]
+            # -sum_j log sigma = +d/2 log(inv_var)  (sigma uniform)
+            bias = (-0.5 * d * reference.LOG_2PI
+                    + 0.5 * d * math.log(inv_var)
+                    - 0.5 * inv_var * (m * m).sum(dim=1))        # [P]
+            return _GMMScoreUniform.apply(feat, A, bias.contiguous(),
+                                          -0.5 * inv_var, apply_exp)
     Wt, bias = gmm_expand_params(means2.float(), covs2.float(), eps)
     return _GMMScore.apply(feat, Wt, bias, apply_exp)
 

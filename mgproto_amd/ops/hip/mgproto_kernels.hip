@@ -157,6 +157,240 @@ void gmm_fwd_kernel(const float* __restrict__ x,   // [N, d]
 }
 
 // ---------------------------------------------------------------------------
+// K1 forward, uniform-sigma reduced form (round 2).
+// With FROZEN isotropic covariances (the reference never updates
+// prototype_covs, model.py:151-152; only means move in the M-step) every
+// x^2-column of W is the same constant cuni = -1/(2 sigma^2), so the x^2
+// half of the GEMM collapses to cuni * ||x||^2 per row:
+//     out[n,p] = f(bias[p] + x[n] . A[p] + cuni * rn2[n])
+// HALF the MFMA FLOPs and half the LDS of the general kernel -> 2 blocks/CU
+// at d=64 (the general <128,128,128> tile is LDS-bound at 1 block/CU,
+// profiles/README.md). Exact fp32 numerics; rn2 is computed by the caller
+// (one cheap elementwise pass).
+// ---------------------------------------------------------------------------
+
+template <int BM, int BN, int KMAX>
+__global__ __launch_bounds__(256)
+void gmm_fwd_uni_kernel(const float* __restrict__ x,   // [N, d]
+                        const float* __restrict__ w,   // [P, d] (A rows)
+                        const float* __restrict__ bias,// [P]
+                        const float* __restrict__ rn2, // [N]
+                        float* __restrict__ out,       // [N, P]
+                        int N, int d, int P, float cuni, int apply_exp) {
+    constexpr int KS = KMAX + 4;
+    constexpr int WM = BM / 2;
+    constexpr int WN = BN / 2;
+    constexpr int FM = WM / 16;
+    constexpr int FN = WN / 16;
+
+    __shared__ float lds[(BM + BN) * KS];
+    float* As = lds;
+    float* Bs = lds + BM * KS;
+
+    const int n0 = blockIdx.x * BM;
+    const int p0 = blockIdx.y * BN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+
+    {   // stage A = x rows
+        const int vec_per_row = d / 4;
+        for (int t = tid; t < BM * vec_per_row; t += 256) {
+            const int r = t / vec_per_row;
+            const int c4 = t % vec_per_row;
+            const int n = min(n0 + r, N - 1);
+            const float4 v = reinterpret_cast<const float4*>(x + (long)n * d)[c4];
+            float* dst = As + r * KS + c4 * 4;
+            dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
+        }
+    }
+    {   // stage B: Bs[p - p0][j] = w[p][j]
+        const int vec_per_row = d / 4;
+        for (int t = tid; t < BN * vec_per_row; t += 256) {
+            const int pr = t / vec_per_row;
+            const int c4 = t % vec_per_row;
+            const int p = min(p0 + pr, P - 1);
+            const float4 v = reinterpret_cast<const float4*>(w + (long)p * d)[c4];
+            float* dst = Bs + pr * KS + c4 * 4;
+            dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
+        }
+    }
+    __syncthreads();
+
+    const int wr = (wave >> 1) * WM;
+    const int wc = (wave & 1) * WN;
+    const int lrow = lane & 15;
+    const int kk = lane >> 4;
+
+    f32x4 acc[FM][FN];
+    #pragma unroll
+    for (int i = 0; i < FM; ++i)
+        #pragma unroll
+        for (int j = 0; j < FN; ++j)
+            acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    for (int k0 = 0; k0 < d; k0 += 16) {
+        f32x4 afr[FM], bfr[FN];
+        #pragma unroll
+        for (int i = 0; i < FM; ++i)
+            afr[i] = *reinterpret_cast<const f32x4*>(
+                As + (wr + i * 16 + lrow) * KS + k0 + 4 * kk);
+        #pragma unroll
+        for (int j = 0; j < FN; ++j)
+            bfr[j] = *reinterpret_cast<const f32x4*>(
+                Bs + (wc + j * 16 + lrow) * KS + k0 + 4 * kk);
+        #pragma unroll
+        for (int q = 0; q < 4; ++q)
+            #pragma unroll
+            for (int i = 0; i < FM; ++i)
+                #pragma unroll
+                for (int j = 0; j < FN; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                        afr[i][q], bfr[j][q], acc[i][j], 0, 0, 0);
+    }
+
+    #pragma unroll
+    for (int i = 0; i < FM; ++i) {
+        #pragma unroll
+        for (int j = 0; j < FN; ++j) {
+            const int col = p0 + wc + j * 16 + lrow;
+            if (col >= P) continue;
+            const float b = bias[col];
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = n0 + wr + i * 16 + (kk * 4 + r);
+                if (row >= N) continue;
+                float v = acc[i][j][r] + b + cuni * rn2[row];
+                if (apply_exp) v = __expf(v);
+                out[(long)row * P + col] = v;
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// K1 backward, uniform-sigma reduced form:
+//     gx[n,j] = (G @ A)[n,j] + 2 cuni * x[n,j] * rs[n],   rs = rowsum(G)
+// K (=P) tiled as in the general backward, but the output tile is [BM, d]
+// (half) and the epilogue combines directly from the accumulators — no LDS
+// scratch round-trip.
+// ---------------------------------------------------------------------------
+
+template <int BM, int KMAX>
+__global__ __launch_bounds__(256)
+void gmm_bwd_uni_kernel(const float* __restrict__ g,   // [N, P]
+                        const float* __restrict__ x,   // [N, d]
+                        const float* __restrict__ w,   // [d, P] (A^T)
+                        const float* __restrict__ rs,  // [N]
+                        float* __restrict__ gx,        // [N, d]
+                        int N, int d, int P, float cuni) {
+    constexpr int BK = 32;
+    constexpr int KS = BK + 4;
+    constexpr int BN = KMAX;                        // all d columns
+    constexpr int WM = BM / 2;
+    constexpr int WN = BN / 2;
+    constexpr int FM = WM / 16;
+    constexpr int FN = WN / 16;
+
+    __shared__ float lds[(BM + BN) * KS];
+    float* Gs = lds;                                // [BM][KS]
+    float* Ws = lds + BM * KS;                      // [BN][KS]
+
+    const int n0 = blockIdx.x * BM;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wr = (wave >> 1) * WM;
+    const int wc = (wave & 1) * WN;
+    const int lrow = lane & 15;
+    const int kk = lane >> 4;
+
+    f32x4 acc[FM][FN];
+    #pragma unroll
+    for (int i = 0; i < FM; ++i)
+        #pragma unroll
+        for (int j = 0; j < FN; ++j)
+            acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    for (int p0 = 0; p0 < P; p0 += BK) {
+        for (int t = tid; t < BM * (BK / 4); t += 256) {
+            const int r = t / (BK / 4);
+            const int c4 = t % (BK / 4);
+            const int n = min(n0 + r, N - 1);
+            const int p = p0 + c4 * 4;
+            float4 v = {0.f, 0.f, 0.f, 0.f};
+            if (p + 3 < P) {
+                v = *reinterpret_cast<const float4*>(g + (long)n * P + p);
+            } else {
+                if (p + 0 < P) v.x = g[(long)n * P + p + 0];
+                if (p + 1 < P) v.y = g[(long)n * P + p + 1];
+                if (p + 2 < P) v.z = g[(long)n * P + p + 2];
+                if (p + 3 < P) v.w = g[(long)n * P + p + 3];
+            }
+            float* dst = Gs + r * KS + c4 * 4;
+            dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
+        }
+        for (int t = tid; t < BN * (BK / 4); t += 256) {
+            const int j = t / (BK / 4);
+            const int c4 = t % (BK / 4);
+            const int p = p0 + c4 * 4;
+            float4 v = {0.f, 0.f, 0.f, 0.f};
+            if (p + 3 < P) {
+                v = *reinterpret_cast<const float4*>(w + (long)j * P + p);
+            } else {
+                if (p + 0 < P) v.x = w[(long)j * P + p + 0];
+                if (p + 1 < P) v.y = w[(long)j * P + p + 1];
+                if (p + 2 < P) v.z = w[(long)j * P + p + 2];
+                if (p + 3 < P) v.w = w[(long)j * P + p + 3];
+            }
+            float* dst = Ws + j * KS + c4 * 4;
+            dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
+        }
+        __syncthreads();
+
+        #pragma unroll
+        for (int km = 0; km < BK; km += 16) {
+            f32x4 afr[FM], bfr[FN];
+            #pragma unroll
+            for (int i = 0; i < FM; ++i)
+                afr[i] = *reinterpret_cast<const f32x4*>(
+                    Gs + (wr + i * 16 + lrow) * KS + km + 4 * kk);
+            #pragma unroll
+            for (int j = 0; j < FN; ++j)
+                bfr[j] = *reinterpret_cast<const f32x4*>(
+                    Ws + (wc + j * 16 + lrow) * KS + km + 4 * kk);
+            #pragma unroll
+            for (int q = 0; q < 4; ++q)
+                #pragma unroll
+                for (int i = 0; i < FM; ++i)
+                    #pragma unroll
+                    for (int j = 0; j < FN; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+                            afr[i][q], bfr[j][q], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+    // epilogue: gx = acc + 2 cuni * x * rs  (direct, no LDS staging)
+    #pragma unroll
+    for (int i = 0; i < FM; ++i) {
+        #pragma unroll
+        for (int j = 0; j < FN; ++j) {
+            const int col = wc + j * 16 + lrow;
+            if (col >= d) continue;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int row = n0 + wr + i * 16 + (kk * 4 + r);
+                if (row >= N) continue;
+                const float xv = x[(long)row * d + col];
+                gx[(long)row * d + col] =
+                    acc[i][j][r] + 2.f * cuni * xv * rs[row];
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // K1 backward: gw[N,2d] = G[N,P] @ W^T, then grad_x = gw[:, :d] + 2x*gw[:, d:]
 // K (=P) is large: tiled K loop, BK=32, A=G rows, B=W columns (w[j2][p] with
 // p as K). Output tile = [BM, 2d] (2d <= 256), one block row per BM rows.
@@ -466,6 +700,60 @@ torch::Tensor gmm_bwd(torch::Tensor g, torch::Tensor x, torch::Tensor w) {
     return gx;
 }
 
+torch::Tensor gmm_fwd_uni(torch::Tensor x, torch::Tensor wr,
+                          torch::Tensor bias, torch::Tensor rn2,
+                          double cuni, bool apply_exp) {
+    CHECK_IN(x); CHECK_IN(wr); CHECK_IN(bias); CHECK_IN(rn2);
+    TORCH_CHECK(x.dtype() == torch::kFloat32, "gmm_fwd_uni: fp32 only");
+    const int N = x.size(0), d = x.size(1), P = wr.size(0);
+    TORCH_CHECK(wr.size(1) == d, "wr must be [P, d]");
+    TORCH_CHECK(rn2.size(0) == N, "rn2 must be [N]");
+    TORCH_CHECK(d % 8 == 0 && d <= 128, "d must be multiple of 8, <= 128");
+    auto out = torch::empty({N, P}, x.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    if (d <= 64) {
+        dim3 grid(ceil_div(N, 128), ceil_div(P, 128));
+        hipLaunchKernelGGL((gmm_fwd_uni_kernel<128, 128, 64>), grid, dim3(256),
+                           0, stream, x.data_ptr<float>(),
+                           wr.data_ptr<float>(), bias.data_ptr<float>(),
+                           rn2.data_ptr<float>(), out.data_ptr<float>(),
+                           N, d, P, (float)cuni, (int)apply_exp);
+    } else {
+        dim3 grid(ceil_div(N, 128), ceil_div(P, 128));
+        hipLaunchKernelGGL((gmm_fwd_uni_kernel<128, 128, 128>), grid, dim3(256),
+                           0, stream, x.data_ptr<float>(),
+                           wr.data_ptr<float>(), bias.data_ptr<float>(),
+                           rn2.data_ptr<float>(), out.data_ptr<float>(),
+                           N, d, P, (float)cuni, (int)apply_exp);
+    }
+    return out;
+}
+
+torch::Tensor gmm_bwd_uni(torch::Tensor g, torch::Tensor x, torch::Tensor w,
+                          torch::Tensor rs, double cuni) {
+    CHECK_IN(g); CHECK_IN(x); CHECK_IN(w); CHECK_IN(rs);
+    const int N = x.size(0), d = x.size(1), P = w.size(1);
+    TORCH_CHECK(g.size(0) == N && g.size(1) == P, "g must be [N, P]");
+    TORCH_CHECK(w.size(0) == d, "w must be [d, P]");
+    TORCH_CHECK(P % 4 == 0, "P must be a multiple of 4");
+    auto gx = torch::empty({N, d}, x.options());
+    auto stream = at::hip::getCurrentHIPStream();
+    if (d <= 64) {
+        hipLaunchKernelGGL((gmm_bwd_uni_kernel<128, 64>),
+                           dim3(ceil_div(N, 128)), dim3(256), 0, stream,
+                           g.data_ptr<float>(), x.data_ptr<float>(),
+                           w.data_ptr<float>(), rs.data_ptr<float>(),
+                           gx.data_ptr<float>(), N, d, P, (float)cuni);
+    } else {
+        hipLaunchKernelGGL((gmm_bwd_uni_kernel<64, 128>),
+                           dim3(ceil_div(N, 64)), dim3(256), 0, stream,
+                           g.data_ptr<float>(), x.data_ptr<float>(),
+                           w.data_ptr<float>(), rs.data_ptr<float>(),
+                           gx.data_ptr<float>(), N, d, P, (float)cuni);
+    }
+    return gx;
+}
+
 std::vector<torch::Tensor> topk_hw(torch::Tensor probs, long T) {
     CHECK_IN(probs);
     const int B = probs.size(0), HW = probs.size(1), P = probs.size(2);
@@ -561,6 +849,10 @@ void bank_push(torch::Tensor feats, torch::Tensor labels, torch::Tensor mem,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gmm_fwd", &gmm_fwd, "fused GMM score GEMM forward (MFMA f32)");
     m.def("gmm_bwd", &gmm_bwd, "GMM score GEMM backward wrt features");
+    m.def("gmm_fwd_uni", &gmm_fwd_uni,
+          "uniform-sigma reduced GMM forward: half-K MFMA + cuni*||x||^2");
+    m.def("gmm_bwd_uni", &gmm_bwd_uni,
+          "uniform-sigma reduced GMM backward wrt features");
     m.def("topk_hw", &topk_hw, "per-(b,p) top-T over HW with indices");
     m.def("argmax_hw", &argmax_hw, "per-(b,p) argmax over HW");
     m.def("bn_fwd", &bn_fwd, "fused BatchNorm(+Add)(+ReLU) forward, NHWC bf16");

@@ -74,6 +74,57 @@ def test_gmm_bwd_parity(N, P, d):
         (got - want).abs().max().item()
 
 
+@pytest.mark.parametrize('N,P,d', [(15680, 2000, 64), (62720, 2000, 64),
+                                   (640, 2000, 128), (512, 370, 64)])
+def test_gmm_fwd_uni_parity(N, P, d):
+    """Uniform-sigma reduced kernel (half-K + cuni*||x||^2) vs oracle."""
+    dev = torch.device('cuda')
+    feat, means, covs = make_gmm(N, P, d, dev, seed=7)
+    inv_var = float(1.0 / (covs.flatten()[0] ** 2))
+    A = (means * inv_var).contiguous()
+    bias = (-0.5 * d * math.log(2 * math.pi)
+            + 0.5 * d * math.log(inv_var)
+            - 0.5 * inv_var * (means * means).sum(dim=1)).contiguous()
+    rn2 = (feat * feat).sum(dim=1).contiguous()
+    out = _ext().gmm_fwd_uni(feat, A, bias, rn2, -0.5 * inv_var, True)
+    want = torch.exp(R.gmm_logprob(feat, means, covs))
+    assert torch.allclose(out, want, atol=1e-4, rtol=1e-4), \
+        (out - want).abs().max().item()
+    out_lp = _ext().gmm_fwd_uni(feat, A, bias, rn2, -0.5 * inv_var, False)
+    want_lp = R.gmm_logprob(feat, means, covs)
+    assert torch.allclose(out_lp, want_lp, atol=1e-4, rtol=1e-4)
+
+
+@pytest.mark.parametrize('N,P,d', [(15680, 2000, 64), (640, 2000, 128),
+                                   (1000, 500, 64)])
+def test_gmm_bwd_uni_parity(N, P, d):
+    dev = torch.device('cuda')
+    feat, means, covs = make_gmm(N, P, d, dev, seed=9)
+    inv_var = float(1.0 / (covs.flatten()[0] ** 2))
+    cuni = -0.5 * inv_var
+    A = (means * inv_var).contiguous()
+    g = torch.randn(N, P, device=dev)
+    rs = g.sum(dim=1).contiguous()
+    got = _ext().gmm_bwd_uni(g.contiguous(), feat, A.t().contiguous(),
+                             rs, cuni)
+    want = g @ A + (2.0 * cuni) * feat * rs.unsqueeze(1)
+    assert torch.allclose(got, want, atol=1e-3, rtol=1e-4), \
+        (got - want).abs().max().item()
+
+
+def test_gmm_scores_dispatch_paths():
+    """Frozen-isotropic covs take the reduced path; general diag sigma
+    keeps the full GEMM — both match the oracle."""
+    from mgproto_amd import ops
+    dev = torch.device('cuda')
+    for uniform in (True, False):
+        feat, means, covs = make_gmm(2048, 500, 64, dev, seed=11,
+                                     uniform_sigma=uniform)
+        out = ops.gmm_scores(feat, means, covs, apply_exp=False)
+        want = R.gmm_logprob(feat.cpu(), means.cpu(), covs.cpu())
+        assert torch.allclose(out.cpu(), want, atol=1e-4, rtol=1e-4), uniform
+
+
 @pytest.mark.parametrize('P', [2000, 370])  # 370: exercises bwd padding
 def test_gmm_autograd_end_to_end(P):
     """Through ops.gmm_scores (native path) vs the CPU reference autograd."""
