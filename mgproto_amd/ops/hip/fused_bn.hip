@@ -178,6 +178,42 @@ void bn_partial_sum_kernel(const float* __restrict__ partials,
     sums[e] = ((a[0] + a[1]) + (a[2] + a[3])) + ((a[4] + a[5]) + (a[6] + a[7]));
 }
 
+// stage 2, block-per-element form: with nb up to 1024 rows the thread-per-
+// element kernel leaves mid-size layers (C<=512) with <=256 active threads
+// doing 1024 serial-ish column loads — 30-40 us while the data is 2 MB.
+// One block per element + fixed-tree LDS reduction is deterministic and
+// fully parallel (adjacent blocks read adjacent columns -> L2-coalesced).
+__global__ __launch_bounds__(256)
+void bn_partial_sum_block_kernel(const float* __restrict__ partials,
+                                 float* __restrict__ sums, int nb, int C2) {
+    const int e = blockIdx.x;
+    const int tid = threadIdx.x;
+    float a = 0.f;
+    for (int b = tid; b < nb; b += 256)
+        a += partials[(long)b * C2 + e];
+    __shared__ float lds[256];
+    lds[tid] = a;
+    __syncthreads();
+    #pragma unroll
+    for (int s = 128; s > 0; s >>= 1) {
+        if (tid < s) lds[tid] += lds[tid + s];
+        __syncthreads();
+    }
+    if (tid == 0) sums[e] = lds[0];
+}
+
+// dispatch: many partial rows + few channels -> block-per-element
+static inline void launch_partial_sum(const float* partials, float* sums,
+                                      int nb, int C2, hipStream_t stream) {
+    if (nb >= 64 && C2 <= 4096) {
+        hipLaunchKernelGGL(bn_partial_sum_block_kernel, dim3(C2), dim3(256),
+                           0, stream, partials, sums, nb, C2);
+    } else {
+        hipLaunchKernelGGL(bn_partial_sum_kernel, dim3((C2 + 255) / 256),
+                           dim3(256), 0, stream, partials, sums, nb, C2);
+    }
+}
+
 // finalize: mean/rstd, running-stat update, scale/shift for the apply pass
 __global__ __launch_bounds__(256)
 void bn_finalize_kernel(const float* __restrict__ sums,
@@ -550,9 +586,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x, torch::Tensor weight,
         hipLaunchKernelGGL(bn_stats_kernel, rgrid, dim3(256),
                            0, stream, (const short*)x.data_ptr(),
                            partials.data_ptr<float>(), M, C);
-        hipLaunchKernelGGL(bn_partial_sum_kernel, dim3(cdiv(2 * C, 256)),
-                           dim3(256), 0, stream, partials.data_ptr<float>(),
-                           sums.data_ptr<float>(), nb, 2 * C);
+        launch_partial_sum(partials.data_ptr<float>(),
+                           sums.data_ptr<float>(), nb, 2 * C, stream);
         hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256),
                            0, stream, sums.data_ptr<float>(),
                            weight.data_ptr<float>(), bias.data_ptr<float>(),
@@ -647,9 +682,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
         else if (relu) RED(true, false);
         else RED(false, false);
         #undef RED
-        hipLaunchKernelGGL(bn_partial_sum_kernel, dim3(cdiv(2 * C, 256)),
-                           dim3(256), 0, stream, partials.data_ptr<float>(),
-                           sums.data_ptr<float>(), nb, 2 * C);
+        launch_partial_sum(partials.data_ptr<float>(),
+                           sums.data_ptr<float>(), nb, 2 * C, stream);
         #define BWD_APPLY(RELU_, RES_, MASK_) \
             hipLaunchKernelGGL((bn_bwd_apply_kernel<RELU_, RES_, MASK_>), \
                                agrid, dim3(256), 0, stream, \
@@ -685,9 +719,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor y,
         else if (relu) RED(true, false);
         else RED(false, false);
         #undef RED
-        hipLaunchKernelGGL(bn_partial_sum_kernel, dim3(cdiv(2 * C, 256)),
-                           dim3(256), 0, stream, partials.data_ptr<float>(),
-                           sums.data_ptr<float>(), nb, 2 * C);
+        launch_partial_sum(partials.data_ptr<float>(),
+                           sums.data_ptr<float>(), nb, 2 * C, stream);
         #define EVAL_APPLY(RELU_, RES_, MASK_) \
             hipLaunchKernelGGL((bn_bwd_eval_kernel<RELU_, RES_, MASK_>), \
                                agrid, dim3(256), 0, stream, \
