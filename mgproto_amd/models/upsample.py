@@ -1,0 +1,50 @@
+"""Exact 2x bilinear upsample with a gather-based backward.
+
+``nn.Upsample(scale_factor=2, mode='bilinear', align_corners=False)``
+backward in PyTorch-ROCm is an atomic scatter: 1.69 ms/step on the
+flagship add-on grads ([80,64,28,28] fp32 — profiles/ round-1 kernel
+stats) for ~40 MB of traffic. The HIP kernels (ops/hip
+up2x_fwd/up2x_bwd) reproduce torch's source-index math exactly
+(s = max(i/2 - 0.25, 0)) and implement backward as a fixed 4-tap
+deterministic gather. Drop-in module: falls back to F.interpolate off
+GPU / for unsupported layouts.
+"""
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+
+def _ext():
+    from ..ops import hip_loader
+    try:
+        return hip_loader.load()
+    except Exception:  # noqa: BLE001
+        return None
+
+
+class _Up2x(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        return _ext().up2x_fwd(x)
+
+    @staticmethod
+    def backward(ctx, gout):
+        gout = gout.contiguous(memory_format=torch.channels_last)
+        return _ext().up2x_bwd(gout)
+
+
+class Upsample2x(nn.Module):
+    """scale_factor=2, mode='bilinear', align_corners=False."""
+
+    def forward(self, x):
+        if (x.is_cuda and x.dim() == 4 and x.shape[1] % 4 == 0
+                and x.dtype in (torch.float32, torch.bfloat16)
+                and _ext() is not None):
+            x = x.contiguous(memory_format=torch.channels_last)
+            return _Up2x.apply(x)
+        return F.interpolate(x, scale_factor=2, mode='bilinear',
+                             align_corners=False)
+
+    def extra_repr(self):
+        return 'scale_factor=2, mode=bilinear (HIP gather backward)'
