@@ -91,7 +91,9 @@ class Comm:
             return t
         n = torch.tensor([t.numel()], device=self.device, dtype=torch.int64)
         ns = self.all_gather_fixed(n).flatten().tolist()
-        mx = max(ns)
+        # mx >= 1: a zero-size all_gather_into_tensor is backend-dependent
+        # (all ranks empty happens e.g. when no push candidates exist yet)
+        mx = max(max(ns), 1)
         buf = t.new_zeros(mx)
         buf[:t.numel()] = t
         out = self.all_gather_fixed(buf)
