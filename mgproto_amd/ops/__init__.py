@@ -65,6 +65,9 @@ class _GMMScore(torch.autograd.Function):
     def forward(ctx, feat, Wt, bias, apply_exp):
         # Wt is [P, 2d] (the fwd kernel's B-panel layout); the bwd kernel
         # streams the [2d, P] transpose, materialized once per call (~1 MB)
+        # B=1 push sweeps reach here with a VIEW (permute+reshape of
+        # [1,d,H,W] is expressible without a copy) — kernels need packed rows
+        feat = feat.contiguous()
         d = feat.shape[1]
         ext = _native_or_die() if feat.is_cuda else None
         if ext is not None and (d % 8 != 0 or d > 128):
@@ -117,6 +120,7 @@ class _GMMScoreUniform(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, feat, A, bias, cuni, apply_exp):
+        feat = feat.contiguous()     # B=1 push sweeps pass strided views
         d = feat.shape[1]
         rn2 = (feat * feat).sum(dim=1).contiguous()
         ext = _native_or_die() if feat.is_cuda else None

@@ -158,6 +158,21 @@ def test_upsample2x_module_autograd():
         (x1.grad - x2.grad).abs().max().item()
 
 
+def test_gmm_scores_strided_view_input():
+    """B=1 push sweeps pass a non-copy permute+reshape VIEW as feat
+    (caught by the GPU eval-driver run, round 2): the kernels need packed
+    rows, so dispatch must make it contiguous."""
+    from mgproto_amd import ops
+    dev = torch.device('cuda')
+    _, means, covs = make_gmm(16, 40, 16, dev, seed=13)
+    base = torch.randn(1, 16, 4, 4, device=dev)
+    feat = base.permute(0, 2, 3, 1).reshape(16, 16)    # strided view
+    assert not feat.is_contiguous()
+    out = ops.gmm_scores(feat, means, covs, apply_exp=False)
+    want = R.gmm_logprob(feat.contiguous().cpu(), means.cpu(), covs.cpu())
+    assert torch.allclose(out.cpu(), want, atol=1e-4, rtol=1e-4)
+
+
 def test_gmm_scores_dispatch_paths():
     """Frozen-isotropic covs take the reduced path; general diag sigma
     keeps the full GEMM — both match the oracle."""
