@@ -38,8 +38,10 @@ class Upsample2x(nn.Module):
     """scale_factor=2, mode='bilinear', align_corners=False."""
 
     def forward(self, x):
+        import os
         if (x.is_cuda and x.dim() == 4 and x.shape[1] % 4 == 0
                 and x.dtype in (torch.float32, torch.bfloat16)
+                and os.environ.get('MGPROTO_NO_UP2X') != '1'
                 and _ext() is not None):
             x = x.contiguous(memory_format=torch.channels_last)
             return _Up2x.apply(x)
