@@ -124,7 +124,9 @@ class _GMMScoreUniform(torch.autograd.Function):
         d = feat.shape[1]
         rn2 = (feat * feat).sum(dim=1).contiguous()
         ext = _native_or_die() if feat.is_cuda else None
-        if ext is not None and (d % 8 != 0 or d > 128):
+        # d%16: the reduced kernel's MFMA macro-step is 16 wide over K=d
+        # (the general kernel's K=2d is always a multiple of 16 for d%8)
+        if ext is not None and (d % 16 != 0 or d > 128):
             ext = None
         ctx.used_ext = ext is not None
         if ext is not None:

@@ -335,13 +335,19 @@ void gmm_bwd_uni_kernel(const float* __restrict__ g,   // [N, P]
             const int c4 = t % (BK / 4);
             const int p = p0 + c4 * 4;
             float4 v = {0.f, 0.f, 0.f, 0.f};
-            if (p + 3 < P) {
-                v = *reinterpret_cast<const float4*>(w + (long)j * P + p);
-            } else {
-                if (p + 0 < P) v.x = w[(long)j * P + p + 0];
-                if (p + 1 < P) v.y = w[(long)j * P + p + 1];
-                if (p + 2 < P) v.z = w[(long)j * P + p + 2];
-                if (p + 3 < P) v.w = w[(long)j * P + p + 3];
+            // j may exceed the REAL row count d (BN is the padded tile
+            // width, e.g. d=16 under KMAX=64): zero-fill those LDS rows
+            // instead of reading w out of bounds — their MFMA columns are
+            // discarded by the epilogue's col<d guard
+            if (j < d) {
+                if (p + 3 < P) {
+                    v = *reinterpret_cast<const float4*>(w + (long)j * P + p);
+                } else {
+                    if (p + 0 < P) v.x = w[(long)j * P + p + 0];
+                    if (p + 1 < P) v.y = w[(long)j * P + p + 1];
+                    if (p + 2 < P) v.z = w[(long)j * P + p + 2];
+                    if (p + 3 < P) v.w = w[(long)j * P + p + 3];
+                }
             }
             float* dst = Ws + j * KS + c4 * 4;
             dst[0] = v.x; dst[1] = v.y; dst[2] = v.z; dst[3] = v.w;
@@ -820,7 +826,7 @@ torch::Tensor gmm_fwd_uni(torch::Tensor x, torch::Tensor wr,
     const int N = x.size(0), d = x.size(1), P = wr.size(0);
     TORCH_CHECK(wr.size(1) == d, "wr must be [P, d]");
     TORCH_CHECK(rn2.size(0) == N, "rn2 must be [N]");
-    TORCH_CHECK(d % 8 == 0 && d <= 128, "d must be multiple of 8, <= 128");
+    TORCH_CHECK(d % 16 == 0 && d <= 128, "d must be multiple of 16, <= 128");
     auto out = torch::empty({N, P}, x.options());
     auto stream = at::hip::getCurrentHIPStream();
     if (d <= 64) {

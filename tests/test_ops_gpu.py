@@ -75,7 +75,8 @@ def test_gmm_bwd_parity(N, P, d):
 
 
 @pytest.mark.parametrize('N,P,d', [(15680, 2000, 64), (62720, 2000, 64),
-                                   (640, 2000, 128), (512, 370, 64)])
+                                   (640, 2000, 128), (512, 370, 64),
+                                   (96, 15, 16)])  # d < KMAX tile (graph-test shape)
 def test_gmm_fwd_uni_parity(N, P, d):
     """Uniform-sigma reduced kernel (half-K + cuni*||x||^2) vs oracle."""
     dev = torch.device('cuda')
@@ -96,7 +97,9 @@ def test_gmm_fwd_uni_parity(N, P, d):
 
 
 @pytest.mark.parametrize('N,P,d', [(15680, 2000, 64), (640, 2000, 128),
-                                   (1000, 500, 64)])
+                                   (1000, 500, 64),
+                                   (96, 16, 16)])  # d < KMAX: w staging must
+                                                   # not read past d rows
 def test_gmm_bwd_uni_parity(N, P, d):
     dev = torch.device('cuda')
     feat, means, covs = make_gmm(N, P, d, dev, seed=9)
