@@ -22,8 +22,11 @@ import torch.nn.functional as F
 def _usable(x: torch.Tensor, conv: nn.Conv2d) -> bool:
     # Measured on the flagship step: plain hipBLASLt matmul on these
     # skinny-K shapes is ~35% SLOWER than MIOpen's CK batched-GEMM picks
-    # (1236 vs 1666 img/s, profiles/bench_history_r01.md), so the GEMM
-    # path is opt-in pending TunableOp-tuned GEMMs.
+    # (1236 vs 1666 img/s, profiles/bench_history_r01.md). Round 2 closed
+    # the question: TunableOp-tuned hipBLASLt still loses end-to-end
+    # (1646.9 vs 1994.2 img/s) and per shape CK wins 12 of 14 — a hybrid
+    # would save 0.04 ms of 4.32 (profiles/bench_r02/conv1x1_per_shape.log).
+    # The GEMM path stays opt-in for A/B only.
     if os.environ.get('MGPROTO_GEMM_CONV1X1') != '1':
         return False
     return (x.is_cuda and x.dim() == 4
