@@ -179,6 +179,49 @@ def test_em_adam_state_survives_resume(tmp_path):
     assert not any(k.startswith('_em_') for k in m1.state_dict())
 
 
+# ------------------------------------------------ gmm dispatch equivalence
+
+def test_gmm_uniform_and_general_paths_agree_cpu():
+    """The uniform-sigma reduced path and the general [x,x^2] path are
+    different code (and different kernels on GPU); on identical inputs
+    they must produce the same log-probs and feature grads."""
+    import torch.nn.functional as F
+    from mgproto_amd import ops
+    g = torch.Generator().manual_seed(4)
+    feat = F.normalize(torch.randn(64, 16, generator=g), dim=1)
+    means = F.normalize(torch.rand(10, 16, generator=g), dim=1)
+    covs = torch.full((10, 16), 0.4)
+
+    f1 = feat.clone().requires_grad_(True)
+    out1 = ops.gmm_scores(f1, means, covs, apply_exp=True)   # uniform path
+    go = torch.randn_like(out1)
+    out1.backward(go)
+
+    os.environ['MGPROTO_NO_GMM_UNI'] = '1'
+    try:
+        f2 = feat.clone().requires_grad_(True)
+        out2 = ops.gmm_scores(f2, means, covs, apply_exp=True)
+        out2.backward(go)
+    finally:
+        del os.environ['MGPROTO_NO_GMM_UNI']
+    assert torch.allclose(out1, out2, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(f1.grad, f2.grad, atol=1e-5, rtol=1e-5)
+
+
+def test_gmm_nonuniform_covs_use_general_path():
+    """Per-prototype sigma must NOT be treated as uniform."""
+    import torch.nn.functional as F
+    from mgproto_amd import ops
+    from mgproto_amd.ops import reference as R
+    g = torch.Generator().manual_seed(5)
+    feat = F.normalize(torch.randn(32, 16, generator=g), dim=1)
+    means = F.normalize(torch.rand(6, 16, generator=g), dim=1)
+    covs = 0.3 + 0.5 * torch.rand(6, 16, generator=g)
+    out = ops.gmm_scores(feat, means, covs, apply_exp=False)
+    want = R.gmm_logprob(feat, means, covs)
+    assert torch.allclose(out, want, atol=1e-5, rtol=1e-5)
+
+
 # --------------------------------------- reducer across phase transitions
 
 @pytest.mark.parametrize('world', [2, 4])
