@@ -66,3 +66,15 @@ def test_reference_state_dict_loads_strict(ab_trajs):
     means the interop contract held against actual reference tensors, not
     just our own export."""
     assert ab_trajs is not None
+
+
+def test_push_projection_matches_reference():
+    """The re-designed push (device argmin + deterministic greedy +
+    batched re-forwards) must produce the SAME prototype projections as
+    the reference's host-numpy per-image push on identical models and
+    images (SURVEY hard part #4)."""
+    from tools.ab_push import run_ab_push
+    torch.manual_seed(0)
+    d_means, chosen, by_proto = run_ab_push()
+    assert d_means < 1e-5, d_means
+    assert len(chosen) == 6          # every prototype re-anchored

@@ -55,6 +55,11 @@ def load_reference(path='/root/reference'):
         pml.miners = types.ModuleType('pytorch_metric_learning.miners')
         pml.losses = types.ModuleType('pytorch_metric_learning.losses')
         sys.modules['pytorch_metric_learning'] = pml
+    if 'matplotlib' not in sys.modules:
+        mpl = types.ModuleType('matplotlib')
+        mpl.pyplot = types.ModuleType('matplotlib.pyplot')
+        sys.modules['matplotlib'] = mpl
+        sys.modules['matplotlib.pyplot'] = mpl.pyplot
     if not torch.cuda.is_available():
         # reference hardcodes .cuda() on a few tensors; make it identity
         torch.Tensor.cuda = lambda self, *a, **k: self  # type: ignore[assignment]
@@ -67,6 +72,16 @@ def load_reference(path='/root/reference'):
         sys.path.remove(path)
     _REF = (ref_model, ref_losses)
     return _REF
+
+
+def load_reference_push(path='/root/reference'):
+    """Import the reference push module (needs the same stubs)."""
+    load_reference(path)
+    sys.path.insert(0, path)
+    try:
+        return importlib.import_module('push')
+    finally:
+        sys.path.remove(path)
 
 
 def make_batches(n_steps, batch, C, img, seed=7):
