@@ -78,3 +78,41 @@ def test_push_projection_matches_reference():
     d_means, chosen, by_proto = run_ab_push()
     assert d_means < 1e-5, d_means
     assert len(chosen) == 6          # every prototype re-anchored
+
+
+def test_prune_matches_reference():
+    """prune_prototypes_topM on identical state zeroes the same prior
+    entries as the reference's per-class loop (reference model.py:467-482)."""
+    from tools.ab_reference import load_reference
+    from mgproto_amd.model import construct_MGProto
+    ref_model_mod, _ = load_reference()
+    torch.manual_seed(0)
+    C, K, d = 5, 4, 16
+    ref = ref_model_mod.construct_MGProto(
+        'resnet18', pretrained=False, img_size=64,
+        prototype_shape=(C * K, d, 1, 1), num_classes=C,
+        add_on_layers_type='regular', sz_embedding=8, mem_capacity=8,
+        mine_K=2)
+    torch.manual_seed(3)
+    ours = construct_MGProto(
+        'resnet18', pretrained=False, img_size=64,
+        prototype_shape=(C * K, d, 1, 1), num_classes=C,
+        add_on_layers_type='regular', sz_embedding=8, mem_capacity=8,
+        mine_K=2)
+    # distinct per-prototype priors so top-M selection is decisive
+    with torch.no_grad():
+        pri = torch.rand(C, K)
+        pri = pri / pri.sum(dim=1, keepdim=True)
+        w = torch.zeros(C, C * K)
+        for c in range(C):
+            w[c, c * K:(c + 1) * K] = pri[c]
+        ref.last_layer.weight.data.copy_(w)
+    ours.load_state_dict(ref.state_dict())
+
+    ref.prune_prototypes_topM(top_M=2)
+    ours.prune_prototypes_topM(top_M=2)
+    assert torch.equal(ours.last_layer.weight.data, ref.last_layer.weight.data)
+    assert torch.equal(ours.prototypes_to_keep.cpu(),
+                       ref.prototypes_to_keep.cpu())
+    assert torch.equal(ours.prototypes_to_keep_with_negative.cpu(),
+                       ref.prototypes_to_keep_with_negative.cpu())
