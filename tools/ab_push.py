@@ -78,7 +78,7 @@ def make_image_tree(root, C=3, per_class=4, size=48, seed=0):
     return samples
 
 
-def run_ab_push(img=64, C=3, K=2, d=16, per_class=4):
+def run_ab_push(img=64, C=3, K=2, d=16, per_class=4, device='cpu'):
     # img 64 -> 4x4 latent grid: the reference's blocked compute_log_prob
     # asserts N %% 4 == 0 even for the B=1 pass-2 re-forwards (model.py:260)
     from mgproto_amd.data import transforms as T
@@ -140,6 +140,9 @@ def run_ab_push(img=64, C=3, K=2, d=16, per_class=4):
             add_on_layers_type='regular', sz_embedding=8,
             mem_capacity=8, mine_K=2)
         our_net.load_state_dict(ref_net.state_dict())   # identical weights
+        if device != 'cpu':
+            ref_net = ref_net.to(device)
+            our_net = our_net.to(device)
 
         ref_art = os.path.join(root, '_ref_art')
         os.makedirs(ref_art, exist_ok=True)
@@ -167,12 +170,13 @@ def run_ab_push(img=64, C=3, K=2, d=16, per_class=4):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument('--out', type=str, default='profiles/ab_push.md')
+    ap.add_argument('--device', type=str, default='cpu')
     args = ap.parse_args()
-    d_means, chosen, by_proto = run_ab_push()
+    d_means, chosen, by_proto = run_ab_push(device=args.device)
     lines = ['# Push-projection equivalence A/B', '',
              'Reference push.py vs engine/push.py on identical models '
-             '(reference state_dict loaded strict), same on-disk images, '
-             'CPU fp32.', '',
+             '(reference state_dict loaded strict), same on-disk images. '
+             f'device={args.device}.', '',
              f'- max |Δ prototype_means| after push: **{d_means:.2e}**',
              f'- prototypes re-anchored by our push: {len(chosen)} '
              f'of {max(j for j, *_ in chosen) + 1 if chosen else 0}',
