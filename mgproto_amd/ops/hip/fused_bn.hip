@@ -349,7 +349,34 @@ void bn_bwd_reduce_kernel(const short* __restrict__ dy,
                 mean[i] = save_mean[c8 + i];
                 rstd[i] = save_rstd[c8 + i];
             }
-            for (long m = blockIdx.x; m < M; m += gridDim.x) {
+            // 2-deep unroll (4-6 loads in flight) — same MLP fix as the
+            // small-C path; this branch serves C>=2048 (DenseNet stages)
+            const long st = gridDim.x;
+            long m = blockIdx.x;
+            for (; m + st < M; m += 2 * st) {
+                short8 g[2], xv[2], yv[2];
+                unsigned char mb[2] = {0xff, 0xff};
+                #pragma unroll
+                for (int u = 0; u < 2; ++u) {
+                    const long mm = m + u * st;
+                    g[u] = *reinterpret_cast<const short8*>(dy + mm * C + c8);
+                    xv[u] = *reinterpret_cast<const short8*>(x + mm * C + c8);
+                    if (RELU && MASK) mb[u] = msk[mm * (C / 8) + (c8 >> 3)];
+                    else if (RELU)
+                        yv[u] = *reinterpret_cast<const short8*>(y + mm * C + c8);
+                }
+                #pragma unroll
+                for (int u = 0; u < 2; ++u)
+                    #pragma unroll
+                    for (int i = 0; i < 8; ++i) {
+                        float gf = bf2f(g[u][i]);
+                        if (RELU && MASK) { if (!((mb[u] >> i) & 1)) gf = 0.f; }
+                        else if (RELU && bf2f(yv[u][i]) <= 0.f) gf = 0.f;
+                        const float xhat = (bf2f(xv[u][i]) - mean[i]) * rstd[i];
+                        sd[i] += gf; sx[i] += gf * xhat;
+                    }
+            }
+            for (; m < M; m += st) {
                 const short8 g = *reinterpret_cast<const short8*>(dy + m * C + c8);
                 const short8 xv = *reinterpret_cast<const short8*>(x + m * C + c8);
                 short8 yv;
