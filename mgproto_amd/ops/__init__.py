@@ -248,7 +248,9 @@ def enqueue_candidates(feat: torch.Tensor, top1_idx: torch.Tensor,
                        gt: torch.Tensor, C: int, K: int, HW: int):
     """Per-sample dedup of the GT class's top-1 patches.
 
-    Opt-in HIP path (MGPROTO_HIP_ENQUEUE=1, pending GPU validation);
+    Opt-in HIP path (MGPROTO_HIP_ENQUEUE=1; round-2 validated: unit +
+    in-situ parity green on MI355X, throughput within noise of the
+    default — docs/ENVVARS.md);
     default = the batched torch sort/mask (reference.py), same output."""
     if (feat.is_cuda and os.environ.get('MGPROTO_HIP_ENQUEUE') == '1'
             and gt.shape[0] <= 1024 and K <= 32

@@ -268,7 +268,7 @@ void bn_eval_coeffs_kernel(const float* __restrict__ weight,
 // apply: y = [relu](x*scale + shift [+ res]); RELU/RES are compile-time
 // channel-resident thread mapping (as bn_stats): fixed c8 per thread, row
 // loop — no per-element integer division, coalesced 16B lanes.
-// MASK (EXPERIMENTAL, round-2: MGPROTO_BN_MASK=1): additionally emit one
+// MASK (default-on since round 2; MGPROTO_BN_MASK=0 disables): emit one
 // relu-mask byte per 8-channel chunk so the backward never re-reads y
 // (saves 2 bf16 activation passes of backward traffic).
 template <bool RELU, bool RES, bool MASK = false>

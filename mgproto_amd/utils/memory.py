@@ -79,7 +79,8 @@ class MemoryBank(nn.Module):
 
         if (feature.is_cuda and feature.dtype == torch.float32
                 and os.environ.get('MGPROTO_HIP_ENQUEUE') == '1'):
-            # K5 HIP path (opt-in, pending GPU validation): deterministic
+            # K5 HIP path (opt-in; round-2 parity-validated on MI355X,
+            # throughput-neutral vs this torch path): deterministic
             # class-segregated ring write, same bank state as the torch
             # sort/scan/scatter below
             try:
