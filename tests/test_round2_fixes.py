@@ -293,3 +293,23 @@ def impl_reducer_warm_joint(rank, world):
             assert torch.allclose(p.grad, po.grad, atol=1e-6), \
                 (phase, n, (p.grad - po.grad).abs().max().item())
     return True
+
+
+# ------------------------------------------------ sampler property fuzz
+
+def test_train_sampler_invariants_fuzz():
+    """Random (N, world, epoch): every rank gets the same count, the union
+    covers the dataset, and padding never exceeds world-1 duplicates."""
+    import random
+    rng = random.Random(0)
+    for _ in range(40):
+        n = rng.randint(1, 300)
+        world = rng.choice([2, 3, 4, 5, 8])
+        epoch = rng.randint(0, 10)
+        per_rank = _sampler_epoch_indices(n, world, epoch)
+        lens = {len(ix) for ix in per_rank}
+        assert len(lens) == 1, (n, world, lens)
+        allix = [i for ix in per_rank for i in ix]
+        assert set(allix) == set(range(n)), (n, world)
+        dup = len(allix) - n
+        assert 0 <= dup < world, (n, world, dup)
