@@ -13,7 +13,8 @@ _HERE = os.path.dirname(os.path.abspath(__file__))
 SRC = [os.path.join(_HERE, 'hip', 'mgproto_kernels.hip'),
        os.path.join(_HERE, 'hip', 'fused_bn.hip'),
        os.path.join(_HERE, 'hip', 'em_kernels.hip'),
-       os.path.join(_HERE, 'hip', 'enqueue_kernels.hip')]
+       os.path.join(_HERE, 'hip', 'enqueue_kernels.hip'),
+       os.path.join(_HERE, 'hip', 'gemm1x1_kernels.hip')]
 OUT = os.path.join(_HERE, '_mgproto_hip.so')
 CPU_SRC = [os.path.join(_HERE, 'cpu', 'fastaug.cpp')]
 CPU_OUT = os.path.join(_HERE, '_mgproto_cpu.so')

@@ -1021,6 +1021,10 @@ void bank_push(torch::Tensor feats, torch::Tensor labels, torch::Tensor mem,
                torch::Tensor head, torch::Tensor mem_len,
                int64_t C, int64_t cap);
 
+std::vector<torch::Tensor> gemm1x1_fwd(torch::Tensor x, torch::Tensor w,
+                                       c10::optional<torch::Tensor> bias,
+                                       bool want_bn_partials);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gmm_fwd", &gmm_fwd, "fused GMM score GEMM forward (MFMA f32)");
     m.def("gmm_bwd", &gmm_bwd, "GMM score GEMM backward wrt features");
@@ -1030,6 +1034,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "uniform-sigma reduced GMM backward wrt features");
     m.def("up2x_fwd", &up2x_fwd,
           "exact 2x bilinear upsample forward, NHWC fp32/bf16");
+    m.def("gemm1x1_fwd", &gemm1x1_fwd,
+          "bf16 MFMA GEMM for stride-1 1x1 convs (+optional BN partials)");
     m.def("up2x_bwd", &up2x_bwd,
           "exact 2x bilinear upsample backward (4-tap gather, no atomics)");
     m.def("topk_hw", &topk_hw, "per-(b,p) top-T over HW with indices");
