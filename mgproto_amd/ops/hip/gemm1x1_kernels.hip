@@ -206,8 +206,10 @@ std::vector<torch::Tensor> gemm1x1_fwd(torch::Tensor x, torch::Tensor w,
     const int BM = wide ? 256 : 128;
     const int BN = wide ? 64 : 128;
     dim3 grid((unsigned)((M + BM - 1) / BM), (unsigned)((N + BN - 1) / BN));
+    // zeros, not empty: each block writes only ITS column tile of its
+    // partial row; the other columns must contribute 0 to the merge
     auto partials = want_bn_partials
-        ? torch::empty({(long)grid.x * grid.y, 2L * N},
+        ? torch::zeros({(long)grid.x * grid.y, 2L * N},
                        x.options().dtype(torch::kFloat32))
         : torch::empty({0}, x.options().dtype(torch::kFloat32));
     float* pp = want_bn_partials ? partials.data_ptr<float>() : nullptr;
