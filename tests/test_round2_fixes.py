@@ -313,3 +313,21 @@ def test_train_sampler_invariants_fuzz():
         assert set(allix) == set(range(n)), (n, world)
         dup = len(allix) - n
         assert 0 <= dup < world, (n, world, dup)
+
+
+def test_faithful_aug_reachable_from_settings(tmp_path):
+    """ADVICE low: cfg.fast_augment=False must select the reference's
+    4-pass chain in the real image loaders."""
+    from PIL import Image
+    d = tmp_path / 'class_0'
+    d.mkdir()
+    Image.new('RGB', (40, 40), (10, 20, 30)).save(d / 'a.jpg')
+    from mgproto_amd.settings import Settings
+    from mgproto_amd.data.loaders import build_image_loaders
+    from mgproto_amd.data import transforms as T
+    for fast, expect in ((True, T.FusedTrainTransform), (False, T.Compose)):
+        cfg = Settings(img_size=32, num_workers=0, train_batch_size=2,
+                       fast_augment=fast)
+        cfg.train_dir = cfg.train_push_dir = cfg.test_dir = str(tmp_path)
+        train_loader, *_ = build_image_loaders(cfg)
+        assert isinstance(train_loader.dataset.transform, expect), fast
