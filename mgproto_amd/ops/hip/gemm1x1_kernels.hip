@@ -72,26 +72,53 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
         for (int j = 0; j < FN; ++j)
             acc[i][j] = (f32x4g){0.f, 0.f, 0.f, 0.f};
 
-    for (int k0 = 0; k0 < K; k0 += BK) {
-        // stage A rows [BM x BK] and B rows [BN x BK]: short8 loads
-        {
-            constexpr int vec = BK / 8;            // short8 per row
-            for (int t = tid; t < BM * vec; t += 256) {
-                const int r = t / vec, c8 = t % vec;
-                const int m = min(n0 + r, M - 1);
-                const short8 v = *reinterpret_cast<const short8*>(
-                    x + (long)m * K + k0 + c8 * 8);
-                *reinterpret_cast<short8*>(As + r * KS + c8 * 8) = v;
-            }
-            for (int t = tid; t < BN * vec; t += 256) {
-                const int r = t / vec, c8 = t % vec;
-                const int n = min(p0 + r, N - 1);
-                const short8 v = *reinterpret_cast<const short8*>(
-                    w + (long)n * K + k0 + c8 * 8);
-                *reinterpret_cast<short8*>(Bs + r * KS + c8 * 8) = v;
-            }
+    // Register-buffered double buffering (v1.5): the NEXT K-tile's global
+    // loads issue before this tile's MFMAs, so HBM latency hides behind
+    // compute; the registers flush to LDS between tiles. No extra LDS.
+    constexpr int VEC = BK / 8;                    // short8 per row
+    constexpr int A_P = BM * VEC / 256;            // pieces per thread
+    constexpr int B_P = BN * VEC / 256;
+
+    short8 ra[A_P], rb[B_P];
+    auto load_tile = [&](int k0) {
+        #pragma unroll
+        for (int p = 0; p < A_P; ++p) {
+            const int t = tid + p * 256;
+            const int r = t / VEC, c8 = t % VEC;
+            const int m = min(n0 + r, M - 1);
+            ra[p] = *reinterpret_cast<const short8*>(
+                x + (long)m * K + k0 + c8 * 8);
         }
-        __syncthreads();
+        #pragma unroll
+        for (int p = 0; p < B_P; ++p) {
+            const int t = tid + p * 256;
+            const int r = t / VEC, c8 = t % VEC;
+            const int n = min(p0 + r, N - 1);
+            rb[p] = *reinterpret_cast<const short8*>(
+                w + (long)n * K + k0 + c8 * 8);
+        }
+    };
+    auto flush_tile = [&]() {
+        #pragma unroll
+        for (int p = 0; p < A_P; ++p) {
+            const int t = tid + p * 256;
+            *reinterpret_cast<short8*>(As + (t / VEC) * KS + (t % VEC) * 8)
+                = ra[p];
+        }
+        #pragma unroll
+        for (int p = 0; p < B_P; ++p) {
+            const int t = tid + p * 256;
+            *reinterpret_cast<short8*>(Bs + (t / VEC) * KS + (t % VEC) * 8)
+                = rb[p];
+        }
+    };
+
+    load_tile(0);
+    flush_tile();
+    __syncthreads();
+    for (int k0 = 0; k0 < K; k0 += BK) {
+        const bool more = k0 + BK < K;
+        if (more) load_tile(k0 + BK);              // overlaps the MFMAs below
 
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {      // two 32-k MFMA steps
@@ -111,7 +138,11 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         afr[i], bfr[j], acc[i][j], 0, 0, 0);
         }
-        __syncthreads();
+        __syncthreads();                           // LDS reads done
+        if (more) {
+            flush_tile();
+            __syncthreads();                       // LDS writes visible
+        }
     }
 
     // epilogue: bias, bf16 store, optional per-block BN partials
