@@ -36,7 +36,7 @@ static __device__ __forceinline__ short f2bf_g(float f) {
 // BM x BN block tile, BK=64 K-tile, 256 threads = 4 waves in a 2x2 wave
 // grid; KS = BK + 8 shorts keeps short8 row reads 16B-aligned and the
 // 16-lane b128 read phases bank-clean (stride 144 B).
-template <int BM, int BN>
+template <int BM, int BN, bool DBUF>
 __global__ __launch_bounds__(256)
 void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
                         const short* __restrict__ w,   // [N, K] bf16
@@ -113,12 +113,14 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
         }
     };
 
+    // DBUF pays only when there are many K-tiles to overlap (measured:
+    // wins at K>=1024, loses below — the spare registers cost occupancy)
     load_tile(0);
     flush_tile();
     __syncthreads();
     for (int k0 = 0; k0 < K; k0 += BK) {
         const bool more = k0 + BK < K;
-        if (more) load_tile(k0 + BK);              // overlaps the MFMAs below
+        if (DBUF && more) load_tile(k0 + BK);      // overlaps the MFMAs below
 
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {      // two 32-k MFMA steps
@@ -140,6 +142,7 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
         }
         __syncthreads();                           // LDS reads done
         if (more) {
+            if (!DBUF) load_tile(k0 + BK);         // plain staged path
             flush_tile();
             __syncthreads();                       // LDS writes visible
         }
@@ -244,16 +247,16 @@ std::vector<torch::Tensor> gemm1x1_fwd(torch::Tensor x, torch::Tensor w,
                        x.options().dtype(torch::kFloat32))
         : torch::empty({0}, x.options().dtype(torch::kFloat32));
     float* pp = want_bn_partials ? partials.data_ptr<float>() : nullptr;
-    if (wide) {
-        hipLaunchKernelGGL((gemm1x1_fwd_kernel<256, 64>), grid, dim3(256), 0,
-                           stream, (const short*)x.data_ptr(),
-                           (const short*)w.data_ptr(), bias_ptr,
-                           (short*)y.data_ptr(), pp, (int)M, K, N);
-    } else {
-        hipLaunchKernelGGL((gemm1x1_fwd_kernel<128, 128>), grid, dim3(256), 0,
-                           stream, (const short*)x.data_ptr(),
-                           (const short*)w.data_ptr(), bias_ptr,
-                           (short*)y.data_ptr(), pp, (int)M, K, N);
-    }
+    const bool dbuf = K >= 1024;   // measured crossover (gemm1x1_hip logs)
+    #define LAUNCH_G(BM_, BN_, DB_) \
+        hipLaunchKernelGGL((gemm1x1_fwd_kernel<BM_, BN_, DB_>), grid, \
+                           dim3(256), 0, stream, (const short*)x.data_ptr(), \
+                           (const short*)w.data_ptr(), bias_ptr, \
+                           (short*)y.data_ptr(), pp, (int)M, K, N)
+    if (wide && dbuf) LAUNCH_G(256, 64, true);
+    else if (wide) LAUNCH_G(256, 64, false);
+    else if (dbuf) LAUNCH_G(128, 128, true);
+    else LAUNCH_G(128, 128, false);
+    #undef LAUNCH_G
     return {y, partials};
 }
