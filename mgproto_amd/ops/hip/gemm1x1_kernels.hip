@@ -36,7 +36,12 @@ static __device__ __forceinline__ short f2bf_g(float f) {
 // BM x BN block tile, BK=64 K-tile, 256 threads = 4 waves in a 2x2 wave
 // grid; KS = BK + 8 shorts keeps short8 row reads 16B-aligned and the
 // 16-lane b128 read phases bank-clean (stride 144 B).
-template <int BM, int BN, bool DBUF>
+// MODE 0: plain staged; 1: register double-buffer (K>=1024 winner);
+// 2: global_load_lds direct staging into an XOR-swizzled LINEAR layout
+// (the LDS-DMA writes wave-uniform-base + lane*16, so padding is
+// impossible — the swizzle keeps ds_read_b128 phases bank-clean and the
+// PER-LANE GLOBAL address carries it, per the CDNA4 guide's caveat).
+template <int BM, int BN, int MODE>
 __global__ __launch_bounds__(256)
 void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
                         const short* __restrict__ w,   // [N, K] bf16
@@ -45,7 +50,9 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
                         float* __restrict__ bn_partial,// [nblk, 2N] or null
                         int M, int K, int N) {
     constexpr int BK = 64;
-    constexpr int KS = BK + 8;
+    constexpr bool GLDS = MODE == 2;
+    constexpr bool DBUF = MODE == 1;
+    constexpr int KS = GLDS ? BK : BK + 8;   // DMA needs the linear layout
     constexpr int WM = BM / 2;
     constexpr int WN = BN / 2;
     constexpr int FM = WM / 16;
@@ -113,10 +120,52 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
         }
     };
 
+    // GLDS: each wave issues 1KB direct global->LDS copies; lane i of a
+    // call covers LDS piece base+i, and its GLOBAL address is pre-swizzled
+    // (chunk c8 = slot ^ (r & 7)) so reads stay conflict-free without pads
+    auto glds_tile = [&](int k0) {
+        constexpr int A_CALLS = BM / 8;            // 64 16B-pieces per call
+        constexpr int B_CALLS = BN / 8;
+        constexpr int A_PW = A_CALLS / 4;          // calls per wave
+        constexpr int B_PW = B_CALLS / 4;
+        #pragma unroll
+        for (int q = 0; q < A_PW; ++q) {
+            const int call = wave * A_PW + q;
+            const int piece = call * 64 + lane;
+            const int r = piece >> 3;
+            const int c8 = (piece & 7) ^ (r & 7);
+            const long m = min(n0 + r, M - 1);
+            const short* g = x + (long)m * K + k0 + c8 * 8;
+            auto* g1 = (const __attribute__((address_space(1))) void*)(
+                reinterpret_cast<uintptr_t>(g));
+            auto* l3 = (__attribute__((address_space(3))) void*)(
+                reinterpret_cast<uintptr_t>(As + call * 512));
+            __builtin_amdgcn_global_load_lds(g1, l3, 16, 0, 0);
+        }
+        #pragma unroll
+        for (int q = 0; q < B_PW; ++q) {
+            const int call = wave * B_PW + q;
+            const int piece = call * 64 + lane;
+            const int r = piece >> 3;
+            const int c8 = (piece & 7) ^ (r & 7);
+            const int n = min(p0 + r, N - 1);
+            const short* g = w + (long)n * K + k0 + c8 * 8;
+            auto* g1 = (const __attribute__((address_space(1))) void*)(
+                reinterpret_cast<uintptr_t>(g));
+            auto* l3 = (__attribute__((address_space(3))) void*)(
+                reinterpret_cast<uintptr_t>(Bs + call * 512));
+            __builtin_amdgcn_global_load_lds(g1, l3, 16, 0, 0);
+        }
+    };
+
     // DBUF pays only when there are many K-tiles to overlap (measured:
     // wins at K>=1024, loses below — the spare registers cost occupancy)
-    load_tile(0);
-    flush_tile();
+    if constexpr (GLDS) {
+        glds_tile(0);
+    } else {
+        load_tile(0);
+        flush_tile();
+    }
     __syncthreads();
     for (int k0 = 0; k0 < K; k0 += BK) {
         const bool more = k0 + BK < K;
@@ -124,15 +173,20 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
 
         #pragma unroll
         for (int ks = 0; ks < BK; ks += 32) {      // two 32-k MFMA steps
+            const int qk = (ks >> 3) + kk;         // 16B chunk index 0..7
             short8 afr[FM], bfr[FN];
             #pragma unroll
-            for (int i = 0; i < FM; ++i)
-                afr[i] = *reinterpret_cast<const short8*>(
-                    As + (wr + i * 16 + lrow) * KS + ks + 8 * kk);
+            for (int i = 0; i < FM; ++i) {
+                const int r = wr + i * 16 + lrow;
+                const int sl = GLDS ? (qk ^ (r & 7)) * 8 : ks + 8 * kk;
+                afr[i] = *reinterpret_cast<const short8*>(As + r * KS + sl);
+            }
             #pragma unroll
-            for (int j = 0; j < FN; ++j)
-                bfr[j] = *reinterpret_cast<const short8*>(
-                    Bs + (wc + j * 16 + lrow) * KS + ks + 8 * kk);
+            for (int j = 0; j < FN; ++j) {
+                const int r = wc + j * 16 + lrow;
+                const int sl = GLDS ? (qk ^ (r & 7)) * 8 : ks + 8 * kk;
+                bfr[j] = *reinterpret_cast<const short8*>(Bs + r * KS + sl);
+            }
             #pragma unroll
             for (int i = 0; i < FM; ++i)
                 #pragma unroll
@@ -142,8 +196,12 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
         }
         __syncthreads();                           // LDS reads done
         if (more) {
-            if (!DBUF) load_tile(k0 + BK);         // plain staged path
-            flush_tile();
+            if constexpr (GLDS) {
+                glds_tile(k0 + BK);
+            } else {
+                if (!DBUF) load_tile(k0 + BK);     // plain staged path
+                flush_tile();
+            }
             __syncthreads();                       // LDS writes visible
         }
     }
@@ -218,7 +276,8 @@ void gemm1x1_fwd_kernel(const short* __restrict__ x,   // [M, K] bf16
 
 std::vector<torch::Tensor> gemm1x1_fwd(torch::Tensor x, torch::Tensor w,
                                        c10::optional<torch::Tensor> bias,
-                                       bool want_bn_partials) {
+                                       bool want_bn_partials,
+                                       int64_t mode /* -1 auto, 0/1/2 */) {
     CHECK_G(x); CHECK_G(w);
     TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16,
                 "gemm1x1_fwd: bf16 only");
@@ -247,16 +306,22 @@ std::vector<torch::Tensor> gemm1x1_fwd(torch::Tensor x, torch::Tensor w,
                        x.options().dtype(torch::kFloat32))
         : torch::empty({0}, x.options().dtype(torch::kFloat32));
     float* pp = want_bn_partials ? partials.data_ptr<float>() : nullptr;
-    const bool dbuf = K >= 1024;   // measured crossover (gemm1x1_hip logs)
-    #define LAUNCH_G(BM_, BN_, DB_) \
-        hipLaunchKernelGGL((gemm1x1_fwd_kernel<BM_, BN_, DB_>), grid, \
+    // auto: measured crossover (gemm1x1_hip logs) — reg-dbuf at K>=1024
+    const int m_ = mode >= 0 ? (int)mode : (K >= 1024 ? 1 : 0);
+    #define LAUNCH_G(BM_, BN_, MD_) \
+        hipLaunchKernelGGL((gemm1x1_fwd_kernel<BM_, BN_, MD_>), grid, \
                            dim3(256), 0, stream, (const short*)x.data_ptr(), \
                            (const short*)w.data_ptr(), bias_ptr, \
                            (short*)y.data_ptr(), pp, (int)M, K, N)
-    if (wide && dbuf) LAUNCH_G(256, 64, true);
-    else if (wide) LAUNCH_G(256, 64, false);
-    else if (dbuf) LAUNCH_G(128, 128, true);
-    else LAUNCH_G(128, 128, false);
+    if (wide) {
+        if (m_ == 2) LAUNCH_G(256, 64, 2);
+        else if (m_ == 1) LAUNCH_G(256, 64, 1);
+        else LAUNCH_G(256, 64, 0);
+    } else {
+        if (m_ == 2) LAUNCH_G(128, 128, 2);
+        else if (m_ == 1) LAUNCH_G(128, 128, 1);
+        else LAUNCH_G(128, 128, 0);
+    }
     #undef LAUNCH_G
     return {y, partials};
 }

@@ -1023,7 +1023,8 @@ void bank_push(torch::Tensor feats, torch::Tensor labels, torch::Tensor mem,
 
 std::vector<torch::Tensor> gemm1x1_fwd(torch::Tensor x, torch::Tensor w,
                                        c10::optional<torch::Tensor> bias,
-                                       bool want_bn_partials);
+                                       bool want_bn_partials,
+                                       int64_t mode);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gmm_fwd", &gmm_fwd, "fused GMM score GEMM forward (MFMA f32)");
@@ -1035,7 +1036,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("up2x_fwd", &up2x_fwd,
           "exact 2x bilinear upsample forward, NHWC fp32/bf16");
     m.def("gemm1x1_fwd", &gemm1x1_fwd,
-          "bf16 MFMA GEMM for stride-1 1x1 convs (+optional BN partials)");
+          "bf16 MFMA GEMM for stride-1 1x1 convs (+optional BN partials)",
+          pybind11::arg("x"), pybind11::arg("w"), pybind11::arg("bias"),
+          pybind11::arg("want_bn_partials"), pybind11::arg("mode") = -1);
     m.def("up2x_bwd", &up2x_bwd,
           "exact 2x bilinear upsample backward (4-tap gather, no atomics)");
     m.def("topk_hw", &topk_hw, "per-(b,p) top-T over HW with indices");

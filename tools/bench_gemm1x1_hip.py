@@ -50,15 +50,21 @@ def main():
         t_ck = timeit(lambda: F.conv2d(x4, w4))
         t_me = timeit(lambda: ext.gemm1x1_fwd(x, w, None, False))
         t_bn = timeit(lambda: ext.gemm1x1_fwd(x, w, None, True))
-        # correctness spot-check on the fly
-        y, _ = ext.gemm1x1_fwd(x, w, None, False)
+        t_md = {m: timeit(lambda m=m: ext.gemm1x1_fwd(x, w, None, False, m))
+                for m in (0, 1, 2)}
+        # correctness spot-checks on the fly (auto + the glds mode)
         want = (x.float() @ w.float().t()).bfloat16().float()
+        y, _ = ext.gemm1x1_fwd(x, w, None, False)
         ok = torch.allclose(y.float(), want, rtol=2e-2, atol=1e-2)
+        y2, _ = ext.gemm1x1_fwd(x, w, None, False, 2)
+        ok2 = torch.allclose(y2.float(), want, rtol=2e-2, atol=1e-2)
         tot_ck += t_ck; tot_mine += t_me; tot_mine_bn += t_bn
         gb = (M * K + K * N + M * N) * 2 / 1e9
         print(f'K={K:5d} N={N:5d} M={M:7d}: CK {t_ck:7.3f} ms | '
-              f'mine {t_me:7.3f} ms ({gb/t_me*1e3:5.0f} GB/s) | '
-              f'+BNstats {t_bn:7.3f} ms | parity={"OK" if ok else "FAIL"}',
+              f'auto {t_me:7.3f} ms ({gb/t_me*1e3:5.0f} GB/s) | '
+              f'plain {t_md[0]:7.3f} dbuf {t_md[1]:7.3f} '
+              f'glds {t_md[2]:7.3f} | +BNstats {t_bn:7.3f} | '
+              f'parity={"OK" if ok else "FAIL"}/{"OK" if ok2 else "FAIL"}',
               flush=True)
     print(f'\ntotals: CK {tot_ck:.3f} ms, mine {tot_mine:.3f} ms, '
           f'mine+BNstats {tot_mine_bn:.3f} ms')
