@@ -306,8 +306,11 @@ std::vector<torch::Tensor> gemm1x1_fwd(torch::Tensor x, torch::Tensor w,
                        x.options().dtype(torch::kFloat32))
         : torch::empty({0}, x.options().dtype(torch::kFloat32));
     float* pp = want_bn_partials ? partials.data_ptr<float>() : nullptr;
-    // auto: measured crossover (gemm1x1_hip logs) — reg-dbuf at K>=1024
-    const int m_ = mode >= 0 ? (int)mode : (K >= 1024 ? 1 : 0);
+    // auto = plain staging: the reg-dbuf crossover at K>=1024 did not
+    // reproduce across boxes (plain won everywhere on the second box,
+    // bench_r02/gemm1x1_hip_v2.log) and glds ~= plain within noise; the
+    // modes remain selectable for the round-3 pipeline work
+    const int m_ = mode >= 0 ? (int)mode : 0;
     #define LAUNCH_G(BM_, BN_, MD_) \
         hipLaunchKernelGGL((gemm1x1_fwd_kernel<BM_, BN_, MD_>), grid, \
                            dim3(256), 0, stream, (const short*)x.data_ptr(), \
